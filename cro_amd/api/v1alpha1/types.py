@@ -1,0 +1,232 @@
+"""Typed API objects for the cro.hpsys.ibm.ie.com/v1alpha1 group.
+
+Field-for-field schema parity with the reference CRDs
+(/root/reference/api/v1alpha1/composabilityrequest_types.go:36-80 and
+/root/reference/api/v1alpha1/composableresource_types.go:27-41): same JSON
+field names (``force_detach``, ``allocation_policy``, ``other_spec``,
+``scalarResource``, ...), same enums, cluster scope, and a separate status
+subresource (enforced by the store, see cro_amd/runtime/store.py).
+
+Also defines the minimal cluster objects the operator consumes (Node,
+ResourceSlice, DeviceTaintRule) — enough surface for the reconcile semantics
+without depending on a kubernetes client package.
+"""
+
+from __future__ import annotations
+
+from typing import ClassVar, Dict, List, Optional
+
+from pydantic import BaseModel, ConfigDict, Field
+
+from .. import _schema_validation as sv  # noqa: F401  (re-exported helpers)
+
+GROUP = "cro.hpsys.ibm.ie.com"
+VERSION = "v1alpha1"
+API_VERSION = f"{GROUP}/{VERSION}"
+
+# ComposableResource / ComposabilityRequest state constants.  The reference
+# declares one set (composabilityrequest_types.go:23-30) but its controllers
+# use literal strings; we name the literal-string machine states directly.
+REQUEST_STATES = ("", "NodeAllocating", "Updating", "Running", "Cleaning", "Deleting")
+RESOURCE_STATES = ("", "Attaching", "Online", "Detaching", "Deleting")
+
+
+class _Model(BaseModel):
+    model_config = ConfigDict(populate_by_name=True, extra="forbid")
+
+
+class ObjectMeta(_Model):
+    """Subset of k8s ObjectMeta the operator relies on (cluster-scoped)."""
+
+    name: str = ""
+    generateName: str = ""
+    uid: str = ""
+    resourceVersion: str = ""
+    generation: int = 0
+    creationTimestamp: Optional[str] = None
+    deletionTimestamp: Optional[str] = None
+    labels: Dict[str, str] = Field(default_factory=dict)
+    annotations: Dict[str, str] = Field(default_factory=dict)
+    finalizers: List[str] = Field(default_factory=list)
+
+
+class K8sObject(_Model):
+    """Base for all stored objects: apiVersion/kind/metadata."""
+
+    KIND: ClassVar[str] = ""
+    apiVersion: str = API_VERSION
+    kind: str = ""
+    metadata: ObjectMeta = Field(default_factory=ObjectMeta)
+
+    def model_post_init(self, __context) -> None:
+        if not self.kind:
+            self.kind = self.KIND
+
+    @property
+    def name(self) -> str:
+        return self.metadata.name
+
+
+# ---------------------------------------------------------------------------
+# ComposabilityRequest (user-facing fleet request)
+# ---------------------------------------------------------------------------
+
+
+class NodeSpecRequirements(_Model):
+    """``other_spec`` capacity requirements.
+
+    Parity: NodeSpec, composabilityrequest_types.go:56-64.
+    """
+
+    milli_cpu: int = 0
+    memory: int = 0
+    ephemeral_storage: int = 0
+    allowed_pod_number: int = 0
+
+
+class ScalarResourceDetails(_Model):
+    """Parity: ScalarResourceDetails, composabilityrequest_types.go:40-54."""
+
+    type: str  # "gpu" | "cxlmemory"
+    model: str
+    size: int
+    force_detach: bool = False
+    allocation_policy: str = "samenode"  # "samenode" | "differentnode"
+    target_node: str = ""
+    other_spec: Optional[NodeSpecRequirements] = None
+
+
+class ScalarResourceStatus(_Model):
+    """Parity: ScalarResourceStatus, composabilityrequest_types.go:73-79."""
+
+    state: str = ""
+    device_id: str = ""
+    cdi_device_id: str = ""
+    node_name: str = ""
+    error: str = ""
+
+
+class ComposabilityRequestSpec(_Model):
+    resource: ScalarResourceDetails
+
+
+class ComposabilityRequestStatus(_Model):
+    """Parity: ComposabilityRequestStatus, composabilityrequest_types.go:67-71."""
+
+    state: str = ""
+    error: str = ""
+    resources: Dict[str, ScalarResourceStatus] = Field(default_factory=dict)
+    scalarResource: Optional[ScalarResourceDetails] = None
+
+
+class ComposabilityRequest(K8sObject):
+    KIND: ClassVar[str] = "ComposabilityRequest"
+    spec: Optional[ComposabilityRequestSpec] = None
+    status: ComposabilityRequestStatus = Field(default_factory=ComposabilityRequestStatus)
+
+
+# ---------------------------------------------------------------------------
+# ComposableResource (internal per-device object)
+# ---------------------------------------------------------------------------
+
+
+class ComposableResourceSpec(_Model):
+    """Parity: ComposableResourceSpec, composableresource_types.go:27-33."""
+
+    type: str
+    model: str
+    target_node: str
+    force_detach: bool = False
+
+
+class ComposableResourceStatus(_Model):
+    """Parity: ComposableResourceStatus, composableresource_types.go:36-41."""
+
+    state: str = ""
+    error: str = ""
+    device_id: str = ""
+    cdi_device_id: str = ""
+
+
+class ComposableResource(K8sObject):
+    KIND: ClassVar[str] = "ComposableResource"
+    spec: Optional[ComposableResourceSpec] = None
+    status: ComposableResourceStatus = Field(default_factory=ComposableResourceStatus)
+
+
+# ---------------------------------------------------------------------------
+# Cluster objects the operator consumes (minimal faithful surface)
+# ---------------------------------------------------------------------------
+
+
+class NodeCapacity(_Model):
+    """Allocatable capacity relevant to other_spec admission (nodes.go:78-117)."""
+
+    milli_cpu: int = 0
+    memory: int = 0
+    ephemeral_storage: int = 0
+    allowed_pod_number: int = 0
+
+
+class NodeStatus(_Model):
+    allocatable: NodeCapacity = Field(default_factory=NodeCapacity)
+    # amdgpu enumeration surfaced by the node agent (device-id -> render node)
+    provider_id: str = ""
+
+
+class Node(K8sObject):
+    KIND: ClassVar[str] = "Node"
+    apiVersion: str = "v1"
+    status: NodeStatus = Field(default_factory=NodeStatus)
+
+
+class ResourceSliceDevice(_Model):
+    """One device row of a DRA ResourceSlice (uuid attribute is the seam the
+    reference matches visibility on, gpus.go:207-239)."""
+
+    name: str = ""
+    uuid: str = ""
+    model: str = ""
+    node: str = ""
+    attributes: Dict[str, str] = Field(default_factory=dict)
+
+
+class ResourceSliceSpec(_Model):
+    node_name: str = ""
+    driver: str = "gpu.amd.com"
+    pool: str = ""
+    devices: List[ResourceSliceDevice] = Field(default_factory=list)
+
+
+class ResourceSlice(K8sObject):
+    KIND: ClassVar[str] = "ResourceSlice"
+    apiVersion: str = "resource.k8s.io/v1alpha3"
+    spec: ResourceSliceSpec = Field(default_factory=ResourceSliceSpec)
+
+
+class DeviceTaintRuleSpec(_Model):
+    """DRA DeviceTaintRule keyed by device uuid (gpus.go:894-989): NoSchedule
+    taint applied to a composed device while it is being detached."""
+
+    device_uuid: str = ""
+    driver: str = "gpu.amd.com"
+    effect: str = "NoSchedule"
+    reason: str = ""
+
+
+class DeviceTaintRule(K8sObject):
+    KIND: ClassVar[str] = "DeviceTaintRule"
+    apiVersion: str = "resource.k8s.io/v1alpha3"
+    spec: DeviceTaintRuleSpec = Field(default_factory=DeviceTaintRuleSpec)
+
+
+ALL_KINDS = {
+    cls.KIND: cls
+    for cls in (
+        ComposabilityRequest,
+        ComposableResource,
+        Node,
+        ResourceSlice,
+        DeviceTaintRule,
+    )
+}

@@ -1,0 +1,309 @@
+"""ComposableResource controller: the per-device 5-state machine.
+
+State-machine parity with the reference
+(composableresource_controller.go:106-434):
+
+    "" → Attaching → Online → Detaching → Deleting
+
+with the amdgpu/CDI node path substituted and one deliberate
+performance departure: where the reference re-queues on fixed 30 s / 3 s
+intervals while waiting for hardware (``:236,298,330,400``), this build uses
+sub-second configurable waits (``ReconcileConfig``) — hardware readiness is
+bounded by the PCIe rescan + amdgpu bind time, not by a poll quantum.  That
+is the p50 attach→CDI-ready win BASELINE.md targets.
+
+Additions over the reference on the Online transition:
+* the CDI spec is written (north-star requirement — the reference delegates
+  device exposure to the NVIDIA plugin stack);
+* an optional gfx950 health probe (HIP MFMA + HBM kernel) validates the
+  composed device actually computes before it is advertised;
+* ``cro_attach_to_ready_seconds`` is observed.
+"""
+
+from __future__ import annotations
+
+import logging
+import time
+from dataclasses import dataclass
+from typing import Dict, Optional
+
+from ..api.v1alpha1.types import ComposableResource
+from ..fabric import Adapter, FabricError, WaitingDeviceAttaching, WaitingDeviceDetaching
+from ..metrics import Metrics
+from ..nodeops import amdgpu, taints
+from ..nodeops.amdgpu import GPULoadsPresent
+from ..nodeops.nodes import node_exists
+from ..runtime.client import Client
+from ..runtime.controller import Reconciler, Result
+from ..runtime.errors import ConflictError, NotFoundError
+
+log = logging.getLogger(__name__)
+
+FINALIZER = "cro.amd.com/finalizer"
+READY_TO_DETACH_LABEL = "cro.amd.com/ready-to-detach-device-id"
+READY_TO_DETACH_CDI_LABEL = "cro.amd.com/ready-to-detach-cdi-device-id"
+MANAGED_BY_LABEL = "app.kubernetes.io/managed-by"
+
+
+@dataclass
+class ReconcileConfig:
+    # wait for device visibility after fabric attach (reference: 30 s)
+    attach_visible_wait: float = 0.05
+    # wait while fabric reports attach/detach in progress (reference: 30 s)
+    fabric_wait: float = 0.25
+    # Online health-check period (reference: 30 s — not latency-critical)
+    online_health_period: float = 30.0
+    # wait for device invisibility during detach (reference: 3 s)
+    detach_invisible_wait: float = 0.05
+
+
+class ComposableResourceReconciler(Reconciler):
+    def __init__(
+        self,
+        client: Client,
+        adapter: Adapter,
+        node_ops: amdgpu.NodeOps,
+        config: Optional[ReconcileConfig] = None,
+    ):
+        self.client = client
+        self.adapter = adapter
+        self.node_ops = node_ops
+        self.config = config or ReconcileConfig()
+        self.metrics = Metrics()
+        # uid → monotonic time of Attaching entry (restart loses the sample,
+        # never corrupts it — the histogram only sees fully observed attaches)
+        self._attach_started: Dict[str, float] = {}
+        self._detach_started: Dict[str, float] = {}
+
+    # -- plumbing ----------------------------------------------------------
+
+    def reconcile(self, name: str) -> Result:
+        try:
+            resource = self.client.get(ComposableResource, name)
+        except NotFoundError:
+            return Result()
+
+        if self._garbage_collect(resource):
+            return Result()
+
+        state = resource.status.state
+        handler = {
+            "": self._handle_none,
+            "Attaching": self._handle_attaching,
+            "Online": self._handle_online,
+            "Detaching": self._handle_detaching,
+            "Deleting": self._handle_deleting,
+        }.get(state)
+        if handler is None:
+            return Result()  # unknown state: parity with reference default (no-op)
+        try:
+            return handler(resource)
+        except ConflictError:
+            raise  # retry with fresh read; no status scribbling on conflicts
+        except Exception as exc:
+            self._set_error(resource, str(exc))
+            raise
+
+    def _set_error(self, resource: ComposableResource, msg: str) -> None:
+        """requeueOnErr parity: persist the failure into .status.error
+        (composableresource_controller.go:436-446)."""
+        try:
+            fresh = self.client.get(ComposableResource, resource.metadata.name)
+            fresh.status.error = msg
+            self.client.update_status(fresh)
+        except Exception:
+            log.debug("could not record error on %s", resource.metadata.name)
+
+    def _garbage_collect(self, resource: ComposableResource) -> bool:
+        """Target node deleted → taint cleanup, Deleting, delete CR
+        (composableresource_controller.go:137-183)."""
+        if not resource.spec or not resource.spec.target_node:
+            return False
+        if node_exists(self.client, resource.spec.target_node):
+            return False
+        if taints.has_device_taint(self.client, resource):
+            taints.delete_device_taint(self.client, resource)
+        did = False
+        if resource.status.state != "Deleting":
+            resource.status.state = "Deleting"
+            resource.status.error = f"target node {resource.spec.target_node} not found"
+            resource = self.client.update_status(resource)
+            did = True
+        if resource.metadata.deletionTimestamp is None:
+            self.client.delete(resource)
+            did = True
+        return did
+
+    # -- states ------------------------------------------------------------
+
+    def _handle_none(self, resource: ComposableResource) -> Result:
+        if FINALIZER not in resource.metadata.finalizers:
+            resource.metadata.finalizers.append(FINALIZER)
+            resource = self.client.update(resource)
+
+        # syncer-created detach CRs carry the device identity in labels
+        # (upstreamsyncer_controller.go:140-165 → :195-202)
+        device_id = resource.metadata.labels.get(READY_TO_DETACH_LABEL, "")
+        if device_id:
+            resource.status.device_id = device_id
+            cdi_id = resource.metadata.labels.get(READY_TO_DETACH_CDI_LABEL, "")
+            if cdi_id:
+                resource.status.cdi_device_id = cdi_id
+
+        self._attach_started[resource.metadata.uid] = time.monotonic()
+        resource.status.state = "Attaching"
+        resource.status.error = ""
+        self.client.update_status(resource)
+        return Result()
+
+    def _handle_attaching(self, resource: ComposableResource) -> Result:
+        if resource.metadata.deletionTimestamp is not None:
+            if resource.status.device_id == "":
+                resource.status.state = "Deleting"
+                self.client.update_status(resource)
+                return Result()
+            elif resource.status.error != "":
+                resource.status.state = "Detaching"
+                self.client.update_status(resource)
+                return Result()
+
+        node = resource.spec.target_node
+        mode = self.adapter.device_resource_type
+
+        self.node_ops.ensure_driver(node)
+
+        if resource.status.device_id == "":
+            t0 = time.monotonic()
+            try:
+                device_id, cdi_device_id = self.adapter.provider.add_resource(resource)
+            except WaitingDeviceAttaching:
+                return Result(requeue_after=self.config.fabric_wait)
+            finally:
+                self.metrics.fabric_request_seconds.labels(
+                    self.adapter.provider.name, "add"
+                ).observe(time.monotonic() - t0)
+            resource.status.error = ""
+            resource.status.device_id = device_id
+            resource.status.cdi_device_id = cdi_device_id
+            resource = self.client.update_status(resource)
+
+        if mode == "DEVICE_PLUGIN":
+            # load check is advisory on attach (reference logs and continues,
+            # composableresource_controller.go:253-256)
+            try:
+                self.node_ops.check_no_loads(node)
+            except GPULoadsPresent as exc:
+                log.warning("gpu loads during attach on %s: %s", node, exc)
+        self.node_ops.refresh_after_attach(node)
+
+        if mode == "DRA":
+            visible = self.node_ops.is_visible_dra(node, resource.status.device_id)
+        else:
+            visible = self.node_ops.is_visible(node, resource.status.device_id)
+        if not visible:
+            return Result(requeue_after=self.config.attach_visible_wait)
+
+        # device enumerable → emit CDI spec and (optionally) verify compute
+        cdi_id = self.node_ops.write_cdi(node, resource.status.device_id)
+        if cdi_id:
+            resource.status.cdi_device_id = cdi_id
+        probe = self.node_ops.health_probe(node, resource.status.device_id)
+        if probe is not None and not probe.get("ok", True):
+            raise FabricError(f"gfx950 health probe failed: {probe}")
+
+        resource.status.state = "Online"
+        resource.status.error = ""
+        self.client.update_status(resource)
+        started = self._attach_started.pop(resource.metadata.uid, None)
+        if started is not None:
+            self.metrics.attach_to_ready_seconds.observe(time.monotonic() - started)
+        self.metrics.devices_online.inc()
+        return Result()
+
+    def _handle_online(self, resource: ComposableResource) -> Result:
+        if resource.metadata.deletionTimestamp is not None:
+            self._detach_started[resource.metadata.uid] = time.monotonic()
+            resource.status.state = "Detaching"
+            self.client.update_status(resource)
+            self.metrics.devices_online.dec()
+            return Result()
+
+        if resource.metadata.labels.get(READY_TO_DETACH_LABEL, ""):
+            # syncer-created CR reached Online bookkeeping: delete to detach
+            # (composableresource_controller.go:310-315)
+            self.client.delete(resource)
+            return Result()
+
+        t0 = time.monotonic()
+        try:
+            self.adapter.provider.check_resource(resource)
+        except FabricError as exc:
+            resource.status.error = str(exc)
+            self.client.update_status(resource)
+            return Result(requeue_after=self.config.online_health_period)
+        finally:
+            self.metrics.fabric_request_seconds.labels(
+                self.adapter.provider.name, "check"
+            ).observe(time.monotonic() - t0)
+        if resource.status.error:
+            resource.status.error = ""
+            self.client.update_status(resource)
+        return Result(requeue_after=self.config.online_health_period)
+
+    def _handle_detaching(self, resource: ComposableResource) -> Result:
+        node = resource.spec.target_node
+        mode = self.adapter.device_resource_type
+
+        if resource.status.device_id != "":
+            if not resource.spec.force_detach:
+                if mode == "DEVICE_PLUGIN":
+                    self.node_ops.check_no_loads(node)  # whole node
+                else:
+                    self.node_ops.check_no_loads(node, resource.status.device_id)
+
+            if mode == "DRA":
+                taints.create_device_taint(self.client, resource)
+
+            self.node_ops.drain(node, resource.status.device_id)
+
+            t0 = time.monotonic()
+            try:
+                self.adapter.provider.remove_resource(resource)
+            except WaitingDeviceDetaching:
+                return Result(requeue_after=self.config.fabric_wait)
+            finally:
+                self.metrics.fabric_request_seconds.labels(
+                    self.adapter.provider.name, "remove"
+                ).observe(time.monotonic() - t0)
+
+            self.node_ops.refresh_after_detach(node)
+
+            if mode == "DRA":
+                visible = self.node_ops.is_visible_dra(node, resource.status.device_id)
+            else:
+                visible = self.node_ops.is_visible(node, resource.status.device_id)
+            if visible:
+                return Result(requeue_after=self.config.detach_invisible_wait)
+
+            self.node_ops.remove_cdi(node, resource.status.device_id)
+
+            if mode == "DRA":
+                taints.delete_device_taint(self.client, resource)
+
+            resource.status.error = ""
+            resource.status.device_id = ""
+            resource.status.cdi_device_id = ""
+            resource = self.client.update_status(resource)
+            started = self._detach_started.pop(resource.metadata.uid, None)
+            if started is not None:
+                self.metrics.detach_seconds.observe(time.monotonic() - started)
+
+        resource.status.state = "Deleting"
+        self.client.update_status(resource)
+        return Result()
+
+    def _handle_deleting(self, resource: ComposableResource) -> Result:
+        if FINALIZER in resource.metadata.finalizers:
+            resource.metadata.finalizers.remove(FINALIZER)
+        self.client.update(resource)
+        return Result()

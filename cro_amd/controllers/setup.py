@@ -1,0 +1,85 @@
+"""Operator wiring: the cmd/main.go:137-201 analog.
+
+``build_manager`` assembles a fully wired Manager: both reconcilers with
+their watch sources and predicates (dual-kind watch with the status-change-
+only predicate, composabilityrequest_controller.go:658-690), the in-process
+admission validator (webhook analog), and optionally the upstream syncer
+runnable.  Callers inject the fabric provider and NodeOps (mock or real) —
+the same wiring serves tests (envtest analog), the bench, and production.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+from ..fabric import Adapter
+from ..nodeops.amdgpu import NodeOps
+from ..runtime.controller import Controller, Source
+from ..runtime.manager import Manager
+from ..runtime.store import InMemoryStore, WatchEvent
+from ..webhook.validator import admission_validator
+from .composabilityrequest import ComposabilityRequestReconciler, RequestReconcileConfig
+from .composableresource import ComposableResourceReconciler, ReconcileConfig
+from .upstreamsyncer import UpstreamSyncer
+
+
+def _status_changed(ev: WatchEvent) -> bool:
+    """UPDATE events pass only when .status changed; CREATE/DELETE are
+    dropped (resourceStatusUpdatePredicate parity, :658-678)."""
+    if ev.type != "MODIFIED":
+        return False
+    if ev.old_object is None:
+        return True
+    return getattr(ev.object, "status", None) != getattr(ev.old_object, "status", None)
+
+
+def build_manager(
+    adapter: Adapter,
+    node_ops: NodeOps,
+    store: Optional[InMemoryStore] = None,
+    resource_config: Optional[ReconcileConfig] = None,
+    request_config: Optional[RequestReconcileConfig] = None,
+    max_concurrent_reconciles: int = 8,
+    enable_webhook: bool = True,
+    syncer_period: Optional[float] = None,
+    syncer_grace: float = 600.0,
+    metrics_port: Optional[int] = None,
+) -> Manager:
+    mgr = Manager(store=store, metrics_port=metrics_port)
+
+    resource_reconciler = ComposableResourceReconciler(
+        mgr.client, adapter, node_ops, resource_config
+    )
+    request_reconciler = ComposabilityRequestReconciler(mgr.client, request_config)
+
+    mgr.add_controller(
+        Controller(
+            "composable_resource",
+            resource_reconciler,
+            sources=[Source(kind="ComposableResource")],
+            max_concurrent_reconciles=max_concurrent_reconciles,
+        )
+    )
+    mgr.add_controller(
+        Controller(
+            "composability_request",
+            request_reconciler,
+            sources=[
+                Source(kind="ComposabilityRequest"),
+                Source(kind="ComposableResource", predicate=_status_changed),
+            ],
+            max_concurrent_reconciles=max_concurrent_reconciles,
+        )
+    )
+
+    if enable_webhook:  # ENABLE_WEBHOOKS!=false parity (cmd/main.go:196-201)
+        mgr.register_admission("ComposabilityRequest", admission_validator(mgr.client))
+
+    if syncer_period is not None:
+        syncer = UpstreamSyncer(mgr.client, adapter, node_ops, grace_period=syncer_grace)
+        mgr.add_runnable(syncer_period, syncer.sync)
+        mgr.syncer = syncer  # exposed for tests
+
+    mgr.resource_reconciler = resource_reconciler
+    mgr.request_reconciler = request_reconciler
+    return mgr

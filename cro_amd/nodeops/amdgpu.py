@@ -1,0 +1,379 @@
+"""amdgpu node operations — the MI355X-native device lifecycle.
+
+Re-design of the reference's nvidia node path (internal/utils/gpus.go,
+SURVEY.md §2.8) for the amdgpu/ROCm stack:
+
+| reference (nvidia)                         | here (amdgpu)                              |
+|--------------------------------------------|--------------------------------------------|
+| modinfo nvidia / ClusterPolicy driver check | /sys/module/amdgpu presence (gpus.go:97-127)|
+| nvidia-smi --query-gpu=gpu_uuid visibility  | KFD topology unique_id scan (gpus.go:207-239)|
+| nvidia-smi --query-compute-apps load check  | /sys/class/kfd/kfd/proc/<pid> scan (gpus.go:241-350)|
+| nvidia-smi drain + sysfs remove + modprobe  | sysfs PCI remove; modprobe -r amdgpu only for last device (gpus.go:352-865)|
+| device-plugin/DCGM daemonset restart        | device-plugin/metrics-exporter restart hooks + ResourceSlice publish (gpus.go:1109-1146)|
+| (no CDI emission)                           | CDI spec write incl. xGMI topology          |
+| (no functional verification)                | HIP health probe on gfx950 (MFMA + HBM), optional |
+
+Two implementations of the same NodeOps surface:
+
+* :class:`AmdNodeOps` — real sysfs/KFD via a NodeExec seam.  Because hot
+  PCI removal on a machine we don't own is destructive, ``destructive=False``
+  redirects the *writes* of the lifecycle (pci remove / rescan / modprobe)
+  into a simulated-detach set while every *read* path (enumeration, loads,
+  CDI) stays real — the mode the single-node GPU bench runs in, where no
+  physical CXL fabric exists to actually compose devices.
+* :class:`MockNodeOps` — pure in-memory double for control-plane tests
+  (the injected-interface analog of the reference's gomonkey MockExecutor,
+  suite_test.go:296-307).
+"""
+
+from __future__ import annotations
+
+import logging
+import threading
+import time
+from typing import Callable, Dict, List, Optional, Set
+
+from ..api.v1alpha1.types import ResourceSlice, ResourceSliceDevice, ResourceSliceSpec
+from .cdi_spec import CDISpecWriter
+from .execs import ExecError, NodeExec
+from .kfd import GPUDevice, enumerate_gpus, enumerate_gpus_amdsmi, gpu_compute_pids
+
+log = logging.getLogger(__name__)
+
+
+class DriverMissing(Exception):
+    pass
+
+
+class GPULoadsPresent(Exception):
+    pass
+
+
+class NodeOps:
+    """Surface the ComposableResource controller programs against."""
+
+    def ensure_driver(self, node: str) -> None:
+        raise NotImplementedError  # pragma: no cover
+
+    def enumerate(self, node: str) -> List[GPUDevice]:
+        raise NotImplementedError  # pragma: no cover
+
+    def is_visible(self, node: str, device_id: str) -> bool:
+        raise NotImplementedError  # pragma: no cover
+
+    def check_no_loads(self, node: str, device_id: Optional[str] = None) -> None:
+        raise NotImplementedError  # pragma: no cover
+
+    def drain(self, node: str, device_id: str) -> None:
+        raise NotImplementedError  # pragma: no cover
+
+    def refresh_after_attach(self, node: str) -> None:
+        raise NotImplementedError  # pragma: no cover
+
+    def refresh_after_detach(self, node: str) -> None:
+        raise NotImplementedError  # pragma: no cover
+
+    def write_cdi(self, node: str, device_id: str) -> str:
+        raise NotImplementedError  # pragma: no cover
+
+    def remove_cdi(self, node: str, device_id: str) -> None:
+        raise NotImplementedError  # pragma: no cover
+
+    def health_probe(self, node: str, device_id: str) -> Optional[dict]:
+        return None  # optional capability
+
+
+class AmdNodeOps(NodeOps):
+    def __init__(
+        self,
+        execer: NodeExec,
+        client=None,
+        cdi_dir: str = "/etc/cdi",
+        destructive: bool = True,
+        initially_detached: Optional[List[str]] = None,
+        restart_hooks: Optional[Dict[str, Callable[[str], None]]] = None,
+        probe_fn: Optional[Callable[[int], dict]] = None,
+        publish_resource_slices: bool = True,
+    ):
+        self.execer = execer
+        self.client = client  # runtime Client for ResourceSlice publication
+        self.cdi = CDISpecWriter(execer, cdi_dir)
+        self.destructive = destructive
+        self.publish_slices = publish_resource_slices
+        # device_ids whose PCI removal is simulated (non-destructive mode)
+        self._sim_detached: Set[str] = set(initially_detached or [])
+        self._sim_lock = threading.Lock()
+        # daemonset-restart analogs, keyed by component name
+        self.restart_hooks = restart_hooks or {}
+        self.probe_fn = probe_fn  # probe_fn(render_minor) -> result dict
+
+    # -- driver ------------------------------------------------------------
+
+    def ensure_driver(self, node: str) -> None:
+        """Host-driver detection: /sys/module/amdgpu (gpus.go:97-127 analog).
+
+        A driver running in a container would surface through a different
+        sysroot on the execer; either way the module directory must exist.
+        """
+        if not self.execer.path_exists(node, "/sys/module/amdgpu"):
+            raise DriverMissing(f"amdgpu kernel module not loaded on node {node}")
+
+    # -- enumeration / visibility -----------------------------------------
+
+    def enumerate(self, node: str) -> List[GPUDevice]:
+        try:
+            gpus = enumerate_gpus(self.execer, node)
+        except ExecError:
+            gpus = enumerate_gpus_amdsmi(self.execer, node)
+        if self.destructive:
+            return gpus
+        with self._sim_lock:
+            return [g for g in gpus if g.device_id not in self._sim_detached]
+
+    def find_gpu(self, node: str, device_id: str) -> Optional[GPUDevice]:
+        for g in self.enumerate(node):
+            if g.device_id == device_id:
+                return g
+        return None
+
+    def is_visible(self, node: str, device_id: str) -> bool:
+        return self.find_gpu(node, device_id) is not None
+
+    def is_visible_dra(self, node: str, device_id: str) -> bool:
+        """DRA visibility: the published ResourceSlice lists the uuid
+        (gpus.go:207-239 ResourceSlice branch)."""
+        if self.client is None:
+            return self.is_visible(node, device_id)
+        for sl in self.client.list(ResourceSlice):
+            if sl.spec.node_name == node:
+                if any(d.uuid == device_id for d in sl.spec.devices):
+                    return True
+        return False
+
+    # -- loads -------------------------------------------------------------
+
+    def check_no_loads(self, node: str, device_id: Optional[str] = None) -> None:
+        gpu_id = None
+        if device_id is not None:
+            gpu = self.find_gpu(node, device_id)
+            if gpu is None:
+                return  # device already gone — nothing can be loading it
+            gpu_id = gpu.gpu_id
+        pids = gpu_compute_pids(self.execer, node, gpu_id)
+        if pids:
+            scope = f"device {device_id}" if device_id else f"node {node}"
+            raise GPULoadsPresent(f"{scope} has active KFD compute processes: {pids}")
+
+    # -- drain / attach refresh -------------------------------------------
+
+    def drain(self, node: str, device_id: str) -> None:
+        """Hot-remove one composed GPU from the node.
+
+        Sequence (gpus.go:352-565 re-designed for amdgpu):
+          1. locate the device (already-gone → idempotent no-op);
+          2. if it is the LAST visible GPU, unload amdgpu first — KFD keeps
+             /dev/kfd open per-process and the module cannot release a lone
+             device while bound;
+          3. write 1 to /sys/bus/pci/devices/<bdf>/remove.
+        No per-GPU persistence daemon exists on AMD (nvidia-persistenced has
+        no analog), and there is no drain-status query — drain progress is
+        tracked in CR status instead.
+        """
+        gpu = self.find_gpu(node, device_id)
+        if gpu is None:
+            return
+        if not self.destructive:
+            with self._sim_lock:
+                self._sim_detached.add(device_id)
+            return
+        remaining = [g for g in self.enumerate(node) if g.device_id != device_id]
+        if not remaining:
+            rc, _, err = self.execer.run(node, ["modprobe", "-r", "amdgpu"], timeout=120)
+            if rc != 0:
+                raise ExecError(f"modprobe -r amdgpu failed: {err}", rc=rc, stderr=err)
+        self.execer.write_file(node, f"/sys/bus/pci/devices/{gpu.pci_bdf}/remove", "1")
+
+    def refresh_after_attach(self, node: str) -> None:
+        """Make a newly composed device enumerable and published.
+
+        PCI rescan binds the amdgpu driver to the hot-added function; then
+        the node's ResourceSlice is re-published (the standalone analog of
+        restarting the DRA kubelet-plugin pod, gpus.go:1109-1146) and the
+        device-plugin/metrics restart hooks fire (nvidia-device-plugin /
+        nvidia-dcgm parity, composableresource_controller.go:257-269).
+        """
+        if self.destructive:
+            self.execer.write_file(node, "/sys/bus/pci/rescan", "1")
+        # non-destructive mode: simulate_compose() already restored the
+        # device into the enumerable set; rescan is a no-op by design
+        for name, hook in self.restart_hooks.items():
+            try:
+                hook(node)
+            except Exception as exc:  # hook failures are non-fatal, like
+                log.warning("restart hook %s failed on %s: %s", name, node, exc)
+        self._publish_slice(node)
+
+    def simulate_compose(self, node: str, device_id: str) -> None:
+        """Non-destructive mode: mark a device as fabric-composed so the next
+        rescan+enumeration sees it (models hot-add on a box with no fabric)."""
+        with self._sim_lock:
+            self._sim_detached.discard(device_id)
+
+    def refresh_after_detach(self, node: str) -> None:
+        for name, hook in self.restart_hooks.items():
+            try:
+                hook(node)
+            except Exception as exc:
+                log.warning("restart hook %s failed on %s: %s", name, node, exc)
+        self._publish_slice(node)
+
+    def _publish_slice(self, node: str) -> None:
+        if self.client is None or not self.publish_slices:
+            return
+        devices = [
+            ResourceSliceDevice(
+                name=f"gpu-{i}",
+                uuid=g.device_id,
+                model="mi355x",
+                node=node,
+                attributes={
+                    "pci-bdf": g.pci_bdf,
+                    "vram-bytes": str(g.vram_bytes),
+                    "xgmi-peers": ",".join(map(str, g.xgmi_peers)),
+                },
+            )
+            for i, g in enumerate(self.enumerate(node))
+        ]
+        name = f"{node}-gpu-pool"
+        existing = self.client.try_get(ResourceSlice, name)
+        if existing is None:
+            sl = ResourceSlice()
+            sl.metadata.name = name
+            sl.spec = ResourceSliceSpec(node_name=node, pool=name, devices=devices)
+            self.client.create(sl)
+        else:
+            existing.spec.devices = devices
+            self.client.update(existing)
+
+    # -- CDI ---------------------------------------------------------------
+
+    def write_cdi(self, node: str, device_id: str) -> str:
+        gpu = self.find_gpu(node, device_id)
+        if gpu is None:
+            raise ExecError(f"device {device_id} not enumerable; cannot write CDI spec")
+        return self.cdi.add_device(node, gpu)
+
+    def remove_cdi(self, node: str, device_id: str) -> None:
+        self.cdi.remove_device(node, device_id)
+
+    # -- health ------------------------------------------------------------
+
+    def health_probe(self, node: str, device_id: str) -> Optional[dict]:
+        if self.probe_fn is None:
+            return None
+        gpu = self.find_gpu(node, device_id)
+        if gpu is None:
+            return None
+        return self.probe_fn(gpu.render_minor)
+
+
+class MockNodeOps(NodeOps):
+    """In-memory NodeOps double for control-plane tests and CPU bench.
+
+    Models per-node visible-device sets with explicit fabric→node latency:
+    a device becomes visible only after ``refresh_after_attach`` (the PCI
+    rescan analog) once the mock fabric composed it.
+    """
+
+    def __init__(self, client=None, attach_visible_delay: float = 0.0):
+        self.client = client
+        self.driver_present: Dict[str, bool] = {}
+        self.composed: Dict[str, Set[str]] = {}  # fabric-composed, pre-rescan
+        self.visible: Dict[str, Set[str]] = {}  # enumerable after rescan
+        self.loads: Dict[str, Set[str]] = {}  # device ids under load ("*" = node)
+        self.cdi_written: Dict[str, Set[str]] = {}
+        self.calls: List[tuple] = []
+        self.attach_visible_delay = attach_visible_delay
+        self._visible_at: Dict[tuple, float] = {}
+        self._lock = threading.RLock()
+
+    # test setup helpers
+    def set_driver(self, node: str, present: bool = True) -> None:
+        self.driver_present[node] = present
+
+    def fabric_composed(self, node: str, device_id: str) -> None:
+        with self._lock:
+            self.composed.setdefault(node, set()).add(device_id)
+
+    def fabric_removed(self, node: str, device_id: str) -> None:
+        with self._lock:
+            self.composed.get(node, set()).discard(device_id)
+
+    def add_load(self, node: str, device_id: str = "*") -> None:
+        with self._lock:
+            self.loads.setdefault(node, set()).add(device_id)
+
+    def clear_loads(self, node: str) -> None:
+        with self._lock:
+            self.loads.pop(node, None)
+
+    # NodeOps surface
+    def ensure_driver(self, node: str) -> None:
+        self.calls.append(("ensure_driver", node))
+        if not self.driver_present.get(node, True):
+            raise DriverMissing(f"amdgpu not loaded on {node}")
+
+    def enumerate(self, node: str):
+        with self._lock:
+            return sorted(self.visible.get(node, set()))
+
+    def is_visible(self, node: str, device_id: str) -> bool:
+        with self._lock:
+            ready_at = self._visible_at.get((node, device_id))
+            if ready_at is not None and time.monotonic() >= ready_at:
+                self.visible.setdefault(node, set()).add(device_id)
+                del self._visible_at[(node, device_id)]
+            return device_id in self.visible.get(node, set())
+
+    def is_visible_dra(self, node: str, device_id: str) -> bool:
+        return self.is_visible(node, device_id)
+
+    def check_no_loads(self, node: str, device_id: Optional[str] = None) -> None:
+        self.calls.append(("check_no_loads", node, device_id))
+        with self._lock:
+            loads = self.loads.get(node, set())
+            if device_id is None:
+                if loads:
+                    raise GPULoadsPresent(f"node {node} busy: {sorted(loads)}")
+            elif device_id in loads or "*" in loads:
+                raise GPULoadsPresent(f"device {device_id} busy")
+
+    def drain(self, node: str, device_id: str) -> None:
+        self.calls.append(("drain", node, device_id))
+        with self._lock:
+            self.visible.get(node, set()).discard(device_id)
+
+    def refresh_after_attach(self, node: str) -> None:
+        self.calls.append(("refresh_after_attach", node))
+        with self._lock:
+            for d in self.composed.get(node, set()) - self.visible.get(node, set()):
+                if self.attach_visible_delay > 0:
+                    self._visible_at.setdefault(
+                        (node, d), time.monotonic() + self.attach_visible_delay
+                    )
+                else:
+                    self.visible.setdefault(node, set()).add(d)
+
+    def refresh_after_detach(self, node: str) -> None:
+        self.calls.append(("refresh_after_detach", node))
+
+    def write_cdi(self, node: str, device_id: str) -> str:
+        self.calls.append(("write_cdi", node, device_id))
+        with self._lock:
+            self.cdi_written.setdefault(node, set()).add(device_id)
+        return f"amd.com/gpu={device_id}"
+
+    def remove_cdi(self, node: str, device_id: str) -> None:
+        self.calls.append(("remove_cdi", node, device_id))
+        with self._lock:
+            self.cdi_written.get(node, set()).discard(device_id)

@@ -1,0 +1,201 @@
+"""KFD topology enumeration — the ROCm-native source of GPU truth.
+
+The reference enumerates devices by forking ``nvidia-smi`` over pod exec and
+scanning /proc/driver/nvidia/gpus (gpus.go:207-239,1362-1448).  The amdgpu
+stack has a faster, fork-free equivalent: the KFD topology tree
+``/sys/class/kfd/kfd/topology/nodes/*`` that the ROCm runtime itself builds
+from.  Reading it costs microseconds, so visibility checks can run at
+event speed instead of CLI speed — one of the levers for beating the
+reference's attach latency.
+
+Per node directory:
+  properties        ``key value`` integer lines; GPUs have simd_count > 0,
+                    unique_id (device serial fuse), drm_render_minor,
+                    vendor_id/device_id, location_id/domain (PCI BDF)
+  gpu_id            KFD-assigned numeric id (used by /sys/class/kfd/kfd/proc)
+  io_links/*/properties  links with type (11 = xGMI) and node_to — the
+                    topology surfaced into CDI specs/ResourceSlices so RCCL
+                    jobs can see the 7-link point-to-point xGMI fabric
+                    (the reference never exposes topology).
+
+``amd-smi``/``rocm-smi`` remain the fallback when KFD sysfs is unreadable.
+"""
+
+from __future__ import annotations
+
+import json
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+from .execs import ExecError, NodeExec
+
+KFD_NODES = "/sys/class/kfd/kfd/topology/nodes"
+KFD_PROC = "/sys/class/kfd/kfd/proc"
+IOLINK_TYPE_XGMI = 11
+
+
+@dataclass
+class GPUDevice:
+    kfd_node: int
+    gpu_id: int
+    device_id: str  # canonical "GPU-<16-hex unique_id>" (or PCI fallback)
+    unique_id: int
+    render_minor: int
+    pci_bdf: str  # "0000:03:00.0"
+    vendor_id: int = 0
+    pci_device_id: int = 0
+    vram_bytes: int = 0
+    gfx_target: str = ""
+    xgmi_peers: List[int] = field(default_factory=list)  # peer kfd node ids
+
+    @property
+    def render_path(self) -> str:
+        return f"/dev/dri/renderD{self.render_minor}"
+
+    @property
+    def card_path(self) -> str:
+        # card minor = render minor - 128 by DRM convention
+        return f"/dev/dri/card{self.render_minor - 128}"
+
+
+def canonical_device_id(unique_id: int, pci_bdf: str) -> str:
+    if unique_id:
+        return f"GPU-{unique_id:016x}"
+    return f"GPU-pci-{pci_bdf}"
+
+
+def _parse_properties(text: str) -> Dict[str, int]:
+    props: Dict[str, int] = {}
+    for line in text.splitlines():
+        parts = line.split()
+        if len(parts) == 2:
+            try:
+                props[parts[0]] = int(parts[1])
+            except ValueError:
+                pass
+    return props
+
+
+def _pci_bdf(props: Dict[str, int]) -> str:
+    # location_id encodes bus/dev/fn: bus = bits 8-15, dev = bits 3-7, fn = 0-2
+    loc = props.get("location_id", 0)
+    domain = props.get("domain", 0)
+    bus = (loc >> 8) & 0xFF
+    dev = (loc >> 3) & 0x1F
+    fn = loc & 0x7
+    return f"{domain:04x}:{bus:02x}:{dev:02x}.{fn:x}"
+
+
+def enumerate_gpus(execer: NodeExec, node: str) -> List[GPUDevice]:
+    """Enumerate GPUs from KFD topology; raises ExecError if KFD is absent."""
+    try:
+        entries = execer.list_dir(node, KFD_NODES)
+    except FileNotFoundError:
+        raise ExecError("KFD topology not present (amdgpu driver not loaded?)")
+
+    gpus: List[GPUDevice] = []
+    for entry in entries:
+        base = f"{KFD_NODES}/{entry}"
+        try:
+            props = _parse_properties(execer.read_file(node, f"{base}/properties"))
+        except FileNotFoundError:
+            continue
+        if props.get("simd_count", 0) <= 0:
+            continue  # CPU node
+        try:
+            gpu_id = int(execer.read_file(node, f"{base}/gpu_id").strip())
+        except (FileNotFoundError, ValueError):
+            gpu_id = 0
+        unique_id = props.get("unique_id", 0)
+        bdf = _pci_bdf(props)
+        dev = GPUDevice(
+            kfd_node=int(entry),
+            gpu_id=gpu_id,
+            device_id=canonical_device_id(unique_id, bdf),
+            unique_id=unique_id,
+            render_minor=props.get("drm_render_minor", 0),
+            pci_bdf=bdf,
+            vendor_id=props.get("vendor_id", 0),
+            pci_device_id=props.get("device_id", 0),
+            gfx_target=str(props.get("gfx_target_version", "")),
+        )
+        # VRAM from mem_banks (heap_type 1/2 = FB public/private)
+        try:
+            for bank in execer.list_dir(node, f"{base}/mem_banks"):
+                bprops = _parse_properties(
+                    execer.read_file(node, f"{base}/mem_banks/{bank}/properties")
+                )
+                if bprops.get("heap_type", 0) in (1, 2):
+                    dev.vram_bytes += bprops.get("size_in_bytes", 0)
+        except FileNotFoundError:
+            pass
+        # xGMI peer links
+        try:
+            for link in execer.list_dir(node, f"{base}/io_links"):
+                lprops = _parse_properties(
+                    execer.read_file(node, f"{base}/io_links/{link}/properties")
+                )
+                if lprops.get("type", 0) == IOLINK_TYPE_XGMI:
+                    dev.xgmi_peers.append(lprops.get("node_to", -1))
+        except FileNotFoundError:
+            pass
+        gpus.append(dev)
+    return gpus
+
+
+def enumerate_gpus_amdsmi(execer: NodeExec, node: str) -> List[GPUDevice]:
+    """Fallback enumeration via ``amd-smi list --json`` (CLI fork — slow path)."""
+    rc, out, err = execer.run(node, ["amd-smi", "list", "--json"])
+    if rc != 0:
+        raise ExecError(f"amd-smi list failed: {err}", rc=rc, stderr=err)
+    data = json.loads(out)
+    gpus: List[GPUDevice] = []
+    rows = data if isinstance(data, list) else data.get("gpus", [])
+    for row in rows:
+        bdf = row.get("bdf", "")
+        uuid = row.get("uuid", "") or ""
+        uid = 0
+        hexpart = uuid.replace("GPU-", "").replace("-", "")
+        try:
+            uid = int(hexpart[:16], 16) if hexpart else 0
+        except ValueError:
+            uid = 0
+        gpus.append(
+            GPUDevice(
+                kfd_node=-1,
+                gpu_id=row.get("gpu", -1),
+                device_id=canonical_device_id(uid, bdf),
+                unique_id=uid,
+                render_minor=0,
+                pci_bdf=bdf,
+            )
+        )
+    return gpus
+
+
+def gpu_compute_pids(execer: NodeExec, node: str, gpu_id: Optional[int] = None) -> List[int]:
+    """PIDs with open KFD compute contexts (the amdgpu-native analog of
+    ``nvidia-smi --query-compute-apps``, gpus.go:241-350).
+
+    /sys/class/kfd/kfd/proc/<pid>/ exists per process with a KFD context;
+    per-GPU attribution uses the vram_<gpu_id> usage files when present.
+    """
+    try:
+        pids = execer.list_dir(node, KFD_PROC)
+    except FileNotFoundError:
+        return []
+    result: List[int] = []
+    for pid in pids:
+        if not pid.isdigit():
+            continue
+        if gpu_id is None:
+            result.append(int(pid))
+            continue
+        try:
+            vram = execer.read_file(node, f"{KFD_PROC}/{pid}/vram_{gpu_id}").strip()
+            if int(vram) > 0:
+                result.append(int(pid))
+        except (FileNotFoundError, ValueError):
+            # no per-GPU attribution available → count it conservatively
+            result.append(int(pid))
+    return result

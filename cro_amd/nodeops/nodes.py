@@ -1,0 +1,79 @@
+"""Node-level helpers: existence, capacity admission, debounced restarts.
+
+Parity with internal/utils/nodes.go:
+
+* :func:`check_node_existed` / :func:`get_all_nodes` (nodes.go:119-144);
+* :func:`check_node_capacity_sufficient` — admission of ``other_spec``
+  CPU/memory/ephemeral/pods requirements against node allocatable
+  (nodes.go:78-117), sized in deployments for 8 × MI355X / 288 GB HBM3E
+  per node;
+* :class:`Debouncer` — the 10-second restart debounce the reference applies
+  to daemonset rollouts and DRA-plugin kills (nodes.go:56-67,
+  gpus.go:1140), generalized so any restart hook can be wrapped.
+"""
+
+from __future__ import annotations
+
+import threading
+import time
+from typing import Callable, Dict, List
+
+from ..api.v1alpha1.types import Node, NodeSpecRequirements
+from ..runtime.client import Client
+from ..runtime.errors import NotFoundError
+
+
+def check_node_existed(client: Client, name: str) -> None:
+    client.get(Node, name)  # raises NotFoundError
+
+
+def node_exists(client: Client, name: str) -> bool:
+    try:
+        client.get(Node, name)
+        return True
+    except NotFoundError:
+        return False
+
+
+def get_all_nodes(client: Client) -> List[Node]:
+    return client.list(Node)
+
+
+def check_node_capacity_sufficient(
+    client: Client, node_name: str, other_spec: NodeSpecRequirements
+) -> bool:
+    node = client.get(Node, node_name)
+    alloc = node.status.allocatable
+    return (
+        alloc.milli_cpu >= other_spec.milli_cpu
+        and alloc.memory >= other_spec.memory
+        and alloc.ephemeral_storage >= other_spec.ephemeral_storage
+        and alloc.allowed_pod_number >= other_spec.allowed_pod_number
+    )
+
+
+class Debouncer:
+    """Per-key debounce: a wrapped call is skipped when the same key fired
+    within ``interval`` seconds (nodes.go:56-67 restartedAt guard)."""
+
+    def __init__(self, interval: float = 10.0):
+        self.interval = interval
+        self._last: Dict[str, float] = {}
+        self._lock = threading.Lock()
+
+    def __call__(self, key: str, fn: Callable[[], None]) -> bool:
+        """Run fn unless debounced; returns True if it ran."""
+        now = time.monotonic()
+        with self._lock:
+            last = self._last.get(key, 0.0)
+            if now - last < self.interval:
+                return False
+            self._last[key] = now
+        fn()
+        return True
+
+    def wrap(self, key_fn: Callable[[str], str], fn: Callable[[str], None]):
+        def wrapped(node: str) -> None:
+            self(key_fn(node), lambda: fn(node))
+
+        return wrapped

@@ -1,0 +1,223 @@
+"""In-process API server: typed object store with k8s-faithful semantics.
+
+This is the coordination bus of the operator (the analog of the kube-apiserver
+the reference talks to, and of the envtest apiserver its tests run against —
+suite_test.go:318-409).  It implements the exact subset of apiserver behavior
+the reconcile semantics depend on:
+
+* optimistic concurrency — update/status-update require the caller's
+  ``resourceVersion`` to match, else :class:`ConflictError` (the reference's
+  state machines rely on conflict-retry for idempotence);
+* a separate **status subresource** — ``update`` never changes status,
+  ``update_status`` never changes spec/metadata (kubebuilder
+  ``+kubebuilder:subresource:status`` on both CRDs);
+* **finalizer** semantics — delete on an object with finalizers sets
+  ``deletionTimestamp`` and fires MODIFIED; the object is removed only when a
+  later update clears the finalizer list (both controllers' Deleting states);
+* **admission** hooks on create/update of the main resource (not status),
+  matching the validating webhook registration
+  (composabilityrequest_webhook.go:49: verbs=create;update, no subresource);
+* **watch** — per-subscriber event queues carrying deep copies, the event
+  source for controllers (watch-driven reconciles are how this build beats
+  the reference's 30 s poll quantum).
+
+Unlike etcd-backed apiservers this store is in-memory; persistence across
+operator restarts is delegated to the real cluster in production deployments
+(cro_amd/runtime/client.py keeps the Client surface identical for both).
+"""
+
+from __future__ import annotations
+
+import itertools
+import queue
+import threading
+import time
+import uuid as uuidlib
+from dataclasses import dataclass
+from typing import Callable, Dict, List, Optional
+
+from ..api import _schema_validation
+from ..api.v1alpha1.types import K8sObject
+from .errors import AdmissionDenied, AlreadyExistsError, ConflictError, NotFoundError
+
+ADDED = "ADDED"
+MODIFIED = "MODIFIED"
+DELETED = "DELETED"
+
+
+@dataclass
+class WatchEvent:
+    type: str  # ADDED | MODIFIED | DELETED
+    object: K8sObject
+    old_object: Optional[K8sObject] = None
+
+
+# admission validator: fn(operation: "CREATE"|"UPDATE", old: obj|None, new: obj)
+AdmissionFn = Callable[[str, Optional[K8sObject], K8sObject], None]
+
+
+def _now_rfc3339() -> str:
+    return time.strftime("%Y-%m-%dT%H:%M:%SZ", time.gmtime())
+
+
+class _Watcher:
+    __slots__ = ("kinds", "queue", "closed")
+
+    def __init__(self, kinds: Optional[List[str]]):
+        self.kinds = set(kinds) if kinds else None
+        self.queue: "queue.Queue[WatchEvent]" = queue.Queue()
+        self.closed = False
+
+    def wants(self, kind: str) -> bool:
+        return self.kinds is None or kind in self.kinds
+
+
+class InMemoryStore:
+    def __init__(self) -> None:
+        self._lock = threading.RLock()
+        self._objects: Dict[str, Dict[str, K8sObject]] = {}
+        self._rv = itertools.count(1)
+        self._watchers: List[_Watcher] = []
+        self._admission: Dict[str, List[AdmissionFn]] = {}
+
+    # -- admission ---------------------------------------------------------
+
+    def register_admission(self, kind: str, fn: AdmissionFn) -> None:
+        self._admission.setdefault(kind, []).append(fn)
+
+    def _admit(self, op: str, old: Optional[K8sObject], new: K8sObject) -> None:
+        for fn in self._admission.get(new.kind, ()):  # failurePolicy=fail
+            fn(op, old, new)
+
+    # -- watch -------------------------------------------------------------
+
+    def watch(self, kinds: Optional[List[str]] = None) -> "queue.Queue[WatchEvent]":
+        w = _Watcher(kinds)
+        with self._lock:
+            self._watchers.append(w)
+        return w.queue
+
+    def _notify(self, ev: WatchEvent) -> None:
+        # deep copies per subscriber so no watcher can mutate shared state
+        for w in self._watchers:
+            if not w.closed and w.wants(ev.object.kind):
+                w.queue.put(
+                    WatchEvent(
+                        ev.type,
+                        ev.object.model_copy(deep=True),
+                        ev.old_object.model_copy(deep=True) if ev.old_object else None,
+                    )
+                )
+
+    # -- CRUD --------------------------------------------------------------
+
+    def create(self, obj: K8sObject) -> K8sObject:
+        obj = obj.model_copy(deep=True)
+        _schema_validation.validate_spec(obj)
+        with self._lock:
+            bucket = self._objects.setdefault(obj.kind, {})
+            if not obj.metadata.name:
+                if obj.metadata.generateName:
+                    obj.metadata.name = obj.metadata.generateName + uuidlib.uuid4().hex[:6]
+                else:
+                    raise ValueError("object has neither name nor generateName")
+            if obj.metadata.name in bucket:
+                raise AlreadyExistsError(f"{obj.kind}/{obj.metadata.name} already exists")
+            self._admit("CREATE", None, obj)
+            obj.metadata.uid = str(uuidlib.uuid4())
+            obj.metadata.resourceVersion = str(next(self._rv))
+            obj.metadata.generation = 1
+            obj.metadata.creationTimestamp = _now_rfc3339()
+            obj.metadata.deletionTimestamp = None
+            bucket[obj.metadata.name] = obj
+            self._notify(WatchEvent(ADDED, obj))
+            return obj.model_copy(deep=True)
+
+    def get(self, kind: str, name: str) -> K8sObject:
+        with self._lock:
+            try:
+                return self._objects[kind][name].model_copy(deep=True)
+            except KeyError:
+                raise NotFoundError(f"{kind}/{name} not found") from None
+
+    def list(self, kind: str, label_selector: Optional[Dict[str, str]] = None) -> List[K8sObject]:
+        with self._lock:
+            items = list(self._objects.get(kind, {}).values())
+            if label_selector:
+                items = [
+                    o
+                    for o in items
+                    if all(o.metadata.labels.get(k) == v for k, v in label_selector.items())
+                ]
+            return [o.model_copy(deep=True) for o in items]
+
+    def update(self, obj: K8sObject) -> K8sObject:
+        """Update metadata+spec; status is preserved from the stored object."""
+        obj = obj.model_copy(deep=True)
+        _schema_validation.validate_spec(obj)
+        with self._lock:
+            stored = self._require(obj.kind, obj.metadata.name)
+            self._check_rv(stored, obj)
+            self._admit("UPDATE", stored, obj)
+            new = obj
+            if hasattr(stored, "status"):
+                new.status = stored.status.model_copy(deep=True)
+            spec_changed = getattr(stored, "spec", None) != getattr(new, "spec", None)
+            new.metadata.uid = stored.metadata.uid
+            new.metadata.creationTimestamp = stored.metadata.creationTimestamp
+            new.metadata.deletionTimestamp = stored.metadata.deletionTimestamp
+            new.metadata.generation = stored.metadata.generation + (1 if spec_changed else 0)
+            new.metadata.resourceVersion = str(next(self._rv))
+            self._objects[obj.kind][obj.metadata.name] = new
+            if new.metadata.deletionTimestamp and not new.metadata.finalizers:
+                return self._finalize_delete(new)
+            self._notify(WatchEvent(MODIFIED, new, stored))
+            return new.model_copy(deep=True)
+
+    def update_status(self, obj: K8sObject) -> K8sObject:
+        """Status-subresource update: only .status is applied."""
+        with self._lock:
+            stored = self._require(obj.kind, obj.metadata.name)
+            self._check_rv(stored, obj)
+            new = stored.model_copy(deep=True)
+            new.status = obj.status.model_copy(deep=True)
+            new.metadata.resourceVersion = str(next(self._rv))
+            self._objects[obj.kind][obj.metadata.name] = new
+            self._notify(WatchEvent(MODIFIED, new, stored))
+            return new.model_copy(deep=True)
+
+    def delete(self, kind: str, name: str) -> None:
+        with self._lock:
+            stored = self._require(kind, name)
+            if stored.metadata.finalizers:
+                if stored.metadata.deletionTimestamp is None:
+                    old = stored.model_copy(deep=True)
+                    stored.metadata.deletionTimestamp = _now_rfc3339()
+                    stored.metadata.resourceVersion = str(next(self._rv))
+                    self._notify(WatchEvent(MODIFIED, stored, old))
+                return
+            self._finalize_delete(stored)
+
+    # -- internals ---------------------------------------------------------
+
+    def _require(self, kind: str, name: str) -> K8sObject:
+        try:
+            return self._objects[kind][name]
+        except KeyError:
+            raise NotFoundError(f"{kind}/{name} not found") from None
+
+    @staticmethod
+    def _check_rv(stored: K8sObject, incoming: K8sObject) -> None:
+        if (
+            incoming.metadata.resourceVersion
+            and incoming.metadata.resourceVersion != stored.metadata.resourceVersion
+        ):
+            raise ConflictError(
+                f"{stored.kind}/{stored.metadata.name}: stale resourceVersion "
+                f"{incoming.metadata.resourceVersion} (stored {stored.metadata.resourceVersion})"
+            )
+
+    def _finalize_delete(self, stored: K8sObject) -> K8sObject:
+        del self._objects[stored.kind][stored.metadata.name]
+        self._notify(WatchEvent(DELETED, stored))
+        return stored.model_copy(deep=True)
