@@ -1,0 +1,205 @@
+"""Single-node embedded operator stack for bench.py, smoke() and GPU tests.
+
+Builds the full operator (manager + both controllers + admission + mock
+fabric) against either:
+
+* the REAL amdgpu node path — KFD enumeration, CDI spec writes, gfx950
+  health probe — in non-destructive lifecycle mode (no physical CXL fabric
+  exists on a bench box, so PCI hot-remove is simulated while every read
+  path and the GPU compute probe are real); this is BASELINE.json config #2
+  measured end to end; or
+* MockNodeOps for CPU-only runs (BASELINE.json config #1: "envtest
+  apiserver + mock fabric backend, CPU-only").
+
+The mock fabric is bound to the real enumerated inventory so the IDs the
+fabric "composes" are devices that actually exist on the node.
+"""
+
+from __future__ import annotations
+
+import os
+import time
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+from .api.v1alpha1.types import (
+    ComposabilityRequest,
+    ComposabilityRequestSpec,
+    Node,
+    ScalarResourceDetails,
+)
+from .controllers import build_manager
+from .controllers.composableresource import ReconcileConfig
+from .controllers.composabilityrequest import RequestReconcileConfig
+from .fabric.adapter import Adapter
+from .fabric.mock import MockFabric, MockFabricConfig
+from .nodeops.amdgpu import AmdNodeOps, MockNodeOps
+from .nodeops.execs import LocalNodeExec
+
+
+@dataclass
+class LocalStack:
+    mgr: object
+    fabric: MockFabric
+    ops: object
+    node_name: str
+    device_ids: List[str] = field(default_factory=list)
+    gpu: bool = False
+
+
+def gpu_available() -> bool:
+    return os.path.exists("/dev/kfd")
+
+
+def build_local_stack(
+    node_name: str = "bench-node",
+    cdi_dir: Optional[str] = None,
+    mode: str = "DRA",
+    use_gpu: Optional[bool] = None,
+    gpu_index: Optional[int] = None,
+    fabric_config: Optional[MockFabricConfig] = None,
+    enable_probe: bool = True,
+    max_concurrent_reconciles: int = 8,
+) -> LocalStack:
+    if use_gpu is None:
+        use_gpu = gpu_available()
+    if cdi_dir is None:
+        cdi_dir = os.path.join(os.environ.get("TMPDIR", "/tmp"), f"cro-cdi-{os.getpid()}")
+
+    if use_gpu:
+        execer = LocalNodeExec()
+        from .nodeops.kfd import enumerate_gpus
+
+        gpus = enumerate_gpus(execer, node_name)
+        if not gpus:
+            raise RuntimeError("no GPUs in KFD topology")
+        if gpu_index is not None:
+            gpus = [gpus[gpu_index % len(gpus)]]
+        device_ids = [g.device_id for g in gpus]
+        probe_fn = None
+        if enable_probe:
+            from .nodeops.probe import probe_fn_for_nodeops
+
+            probe_fn = probe_fn_for_nodeops
+        fabric = MockFabric(
+            config=fabric_config,
+            bind_inventory=[
+                {"device_id": g.device_id, "cdi_device_id": f"amd.com/gpu={g.device_id}", "model": "mi355x"}
+                for g in gpus
+            ],
+        )
+        adapter = Adapter(mode, fabric)
+        mgr = build_manager(
+            adapter,
+            None,  # node_ops installed below (needs mgr.client)
+            resource_config=ReconcileConfig(),
+            request_config=RequestReconcileConfig(),
+            max_concurrent_reconciles=max_concurrent_reconciles,
+        )
+        ops = AmdNodeOps(
+            execer,
+            client=mgr.client,
+            cdi_dir=cdi_dir,
+            destructive=False,
+            initially_detached=device_ids,  # "in the fabric pool, not composed"
+            probe_fn=probe_fn,
+        )
+        mgr.resource_reconciler.node_ops = ops
+
+        # bridge fabric composition → simulated hot-add (a real fabric would
+        # make the device appear on the PCIe bus; here the silicon is already
+        # present, so composition clears the simulated-detached mark)
+        orig_add = fabric.add_resource
+
+        def add_resource(resource):
+            device_id, cdi_id = orig_add(resource)
+            ops.simulate_compose(resource.spec.target_node, device_id)
+            return device_id, cdi_id
+
+        fabric.add_resource = add_resource
+    else:
+        fabric = MockFabric(config=fabric_config, models={"mi355x": 8})
+        adapter = Adapter(mode, fabric)
+        mgr = build_manager(
+            adapter,
+            None,
+            resource_config=ReconcileConfig(),
+            request_config=RequestReconcileConfig(),
+            max_concurrent_reconciles=max_concurrent_reconciles,
+        )
+        ops = MockNodeOps(client=mgr.client)
+        mgr.resource_reconciler.node_ops = ops
+        ops.set_driver(node_name, True)
+        orig_add = fabric.add_resource
+
+        def add_resource(resource):
+            device_id, cdi_id = orig_add(resource)
+            ops.fabric_composed(resource.spec.target_node, device_id)
+            return device_id, cdi_id
+
+        fabric.add_resource = add_resource
+        device_ids = []
+
+    node = Node()
+    node.metadata.name = node_name
+    node.status.allocatable.milli_cpu = 128000
+    node.status.allocatable.memory = 2 << 40
+    node.status.allocatable.allowed_pod_number = 256
+    mgr.client.create(node)
+
+    return LocalStack(
+        mgr=mgr, fabric=fabric, ops=ops, node_name=node_name,
+        device_ids=device_ids, gpu=use_gpu,
+    )
+
+
+def attach_detach_cycle(stack: LocalStack, name: str, size: int = 1, timeout: float = 60.0) -> dict:
+    """One full ComposabilityRequest lifecycle; returns timing samples.
+
+    attach_ms = create → Running (every device Online with CDI written);
+    detach_ms = delete → object gone (device drained + fabric detach done).
+    """
+    mgr = stack.mgr
+    req = ComposabilityRequest(
+        spec=ComposabilityRequestSpec(
+            resource=ScalarResourceDetails(
+                type="gpu", model="mi355x", size=size, target_node=stack.node_name
+            )
+        )
+    )
+    req.metadata.name = name
+
+    t0 = time.monotonic()
+    mgr.client.create(req)
+    ok = mgr.wait_for(
+        lambda: mgr.client.try_get(ComposabilityRequest, name) is not None
+        and mgr.client.get(ComposabilityRequest, name).status.state == "Running",
+        timeout=timeout,
+    )
+    t1 = time.monotonic()
+    if not ok:
+        cur = mgr.client.try_get(ComposabilityRequest, name)
+        raise RuntimeError(
+            f"request {name} did not reach Running in {timeout}s "
+            f"(state={cur.status.state if cur else 'gone'}, "
+            f"err={cur.status.error if cur else ''})"
+        )
+
+    mgr.client.delete(ComposabilityRequest, name)
+    ok = mgr.wait_for(
+        lambda: mgr.client.try_get(ComposabilityRequest, name) is None, timeout=timeout
+    )
+    t2 = time.monotonic()
+    if not ok:
+        raise RuntimeError(f"request {name} did not tear down in {timeout}s")
+    return {"attach_ms": (t1 - t0) * 1e3, "detach_ms": (t2 - t1) * 1e3}
+
+
+def reconcile_count(stack: LocalStack) -> float:
+    """Total reconcile invocations across both controllers (Prometheus)."""
+    total = 0.0
+    for metric in stack.mgr.metrics.reconcile_total.collect():
+        for sample in metric.samples:
+            if sample.name.endswith("_total"):
+                total += sample.value
+    return total

@@ -105,7 +105,7 @@ class AmdNodeOps(NodeOps):
         self._sim_lock = threading.Lock()
         # daemonset-restart analogs, keyed by component name
         self.restart_hooks = restart_hooks or {}
-        self.probe_fn = probe_fn  # probe_fn(render_minor) -> result dict
+        self.probe_fn = probe_fn  # probe_fn(gpu: GPUDevice) -> result dict
 
     # -- driver ------------------------------------------------------------
 
@@ -160,6 +160,14 @@ class AmdNodeOps(NodeOps):
                 return  # device already gone — nothing can be loading it
             gpu_id = gpu.gpu_id
         pids = gpu_compute_pids(self.execer, node, gpu_id)
+        # node-agent self-exemption: when operating on the local node, our own
+        # process holds a KFD context for the health probe — it must not block
+        # the detach it is itself orchestrating
+        import os as _os
+        from .execs import LocalNodeExec as _Local
+
+        if isinstance(self.execer, _Local):
+            pids = [p for p in pids if p != _os.getpid()]
         if pids:
             scope = f"device {device_id}" if device_id else f"node {node}"
             raise GPULoadsPresent(f"{scope} has active KFD compute processes: {pids}")
@@ -274,7 +282,7 @@ class AmdNodeOps(NodeOps):
         gpu = self.find_gpu(node, device_id)
         if gpu is None:
             return None
-        return self.probe_fn(gpu.render_minor)
+        return self.probe_fn(gpu)
 
 
 class MockNodeOps(NodeOps):
