@@ -237,6 +237,29 @@ extern "C" int cro_probe_run(int device, struct CroProbeResult* out) {
   return 0;
 }
 
+// Raw MFMA tile entry for numerics tests: D(16x16) = A(16xK) @ B(Kx16) on
+// the f32 matrix pipe; compared against a PyTorch fp32 reference in
+// tests/test_gpu.py.  K must be a multiple of 4.
+extern "C" int cro_probe_mfma_f32(int device, const float* A, const float* B,
+                                  float* D, int K) {
+  struct CroProbeResult scratch;
+  struct CroProbeResult* out = &scratch;  // reuse CHECK() plumbing
+  memset(out, 0, sizeof(*out));
+  if (K <= 0 || (K & 3) != 0) return -10;
+  CHECK(hipSetDevice(device));
+  float *dA, *dB, *dD;
+  CHECK(hipMalloc(&dA, 16 * K * sizeof(float)));
+  CHECK(hipMalloc(&dB, K * 16 * sizeof(float)));
+  CHECK(hipMalloc(&dD, 256 * sizeof(float)));
+  CHECK(hipMemcpy(dA, A, 16 * K * sizeof(float), hipMemcpyHostToDevice));
+  CHECK(hipMemcpy(dB, B, K * 16 * sizeof(float), hipMemcpyHostToDevice));
+  hipLaunchKernelGGL(mfma_f32_check_kernel, dim3(1), dim3(64), 0, 0, dA, dB, dD, K);
+  CHECK(hipGetLastError());
+  CHECK(hipMemcpy(D, dD, 256 * sizeof(float), hipMemcpyDeviceToHost));
+  (void)hipFree(dA); (void)hipFree(dB); (void)hipFree(dD);
+  return 0;
+}
+
 extern "C" int cro_probe_device_count(void) {
   int count = 0;
   if (hipGetDeviceCount(&count) != hipSuccess) return -1;

@@ -1,0 +1,157 @@
+"""Full-manager integration tests: watch-driven reconciles, concurrent
+worker fan-out, churn and contention — the BASELINE.json config shapes
+(#1 envtest+mock, #3 8-device bulk, #4 scale churn, #5 contention)."""
+
+import pytest
+
+from cro_amd.api.v1alpha1.types import ComposabilityRequest, ComposableResource
+from cro_amd.bench_harness import attach_detach_cycle, build_local_stack
+from cro_amd.fabric.mock import MockFabricConfig
+from cro_amd.runtime.errors import AdmissionDenied
+from tests.conftest import make_request
+
+
+@pytest.fixture
+def stack():
+    s = build_local_stack(node_name="node0", use_gpu=False)
+    s.mgr.start()
+    yield s
+    s.mgr.stop()
+
+
+def running(stack, name):
+    req = stack.mgr.client.try_get(ComposabilityRequest, name)
+    return req is not None and req.status.state == "Running"
+
+
+def test_e2e_single_device(stack):
+    timing = attach_detach_cycle(stack, "r1", size=1)
+    assert timing["attach_ms"] < 5000
+    assert timing["detach_ms"] < 5000
+
+
+def test_e2e_bulk_8_devices(stack):
+    """One CR composing 8 devices through concurrent reconcile fan-out."""
+    req = make_request("bulk", size=8, target_node="node0")
+    stack.mgr.client.create(req)
+    assert stack.mgr.wait_for(lambda: running(stack, "bulk"), timeout=15)
+    got = stack.mgr.client.get(ComposabilityRequest, "bulk")
+    assert len(got.status.resources) == 8
+    ids = {v.device_id for v in got.status.resources.values()}
+    assert len(ids) == 8  # distinct devices
+    assert len(stack.fabric.attached_to("node0")) == 8
+    stack.mgr.client.delete(ComposabilityRequest, "bulk")
+    assert stack.mgr.wait_for(
+        lambda: stack.mgr.client.try_get(ComposabilityRequest, "bulk") is None,
+        timeout=15,
+    )
+    assert stack.fabric.attached_to("node0") == []
+
+
+def test_scale_churn_1_4_8_0(stack):
+    """Scale 1→4→8→0 via spec updates under the admission validator
+    (BASELINE config #4)."""
+    stack.mgr.client.create(make_request("churn", size=1, target_node="node0"))
+    assert stack.mgr.wait_for(lambda: running(stack, "churn"), timeout=10)
+
+    for size in (4, 8):
+        req = stack.mgr.client.get(ComposabilityRequest, "churn")
+        req.spec.resource.size = size
+        stack.mgr.client.update(req)
+        assert stack.mgr.wait_for(
+            lambda: running(stack, "churn")
+            and len(stack.mgr.client.get(ComposabilityRequest, "churn").status.resources) == size,
+            timeout=20,
+        ), f"scale to {size} failed"
+        assert len(stack.fabric.attached_to("node0")) == size
+
+    req = stack.mgr.client.get(ComposabilityRequest, "churn")
+    req.spec.resource.size = 0
+    stack.mgr.client.update(req)
+    assert stack.mgr.wait_for(
+        lambda: running(stack, "churn")
+        and len(stack.mgr.client.get(ComposabilityRequest, "churn").status.resources) == 0,
+        timeout=20,
+    )
+    # child detaches complete asynchronously after the request is Running
+    assert stack.mgr.wait_for(lambda: stack.fabric.attached_to("node0") == [], timeout=20)
+    stack.mgr.client.delete(ComposabilityRequest, "churn")
+    assert stack.mgr.wait_for(
+        lambda: stack.mgr.client.try_get(ComposabilityRequest, "churn") is None, timeout=10
+    )
+
+
+def test_contention_two_requests_one_pool(stack):
+    """Two requests contending for the 8-device pool (BASELINE config #5);
+    the admission rules force distinct (type,model,node) keys, so contention
+    happens at the fabric pool."""
+    from cro_amd.api.v1alpha1.types import Node
+
+    n1 = Node()
+    n1.metadata.name = "node1"
+    stack.mgr.client.create(n1)
+    stack.ops.set_driver("node1", True)
+
+    stack.mgr.client.create(make_request("a", size=5, target_node="node0"))
+    stack.mgr.client.create(make_request("b", size=3, target_node="node1"))
+    assert stack.mgr.wait_for(
+        lambda: running(stack, "a") and running(stack, "b"), timeout=20
+    )
+    assert len(stack.fabric.attached_to("node0")) == 5
+    assert len(stack.fabric.attached_to("node1")) == 3
+    # pool exhausted: a third request must surface an error, not wedge others
+    stack.mgr.client.create(make_request("c", size=1, target_node="node0", model="mi300x"))
+    assert stack.mgr.wait_for(
+        lambda: (
+            stack.mgr.client.try_get(ComposabilityRequest, "c") is not None
+            and any(
+                "no free" in (v.error or "")
+                for v in stack.mgr.client.get(ComposabilityRequest, "c").status.resources.values()
+            )
+        ),
+        timeout=10,
+    )
+    for name in ("a", "b", "c"):
+        stack.mgr.client.delete(ComposabilityRequest, name)
+    assert stack.mgr.wait_for(
+        lambda: all(
+            stack.mgr.client.try_get(ComposabilityRequest, n) is None
+            for n in ("a", "b", "c")
+        ),
+        timeout=20,
+    )
+
+
+def test_webhook_active_through_manager(stack):
+    stack.mgr.client.create(make_request("w1", policy="differentnode"))
+    with pytest.raises(AdmissionDenied):
+        stack.mgr.client.create(make_request("w2", policy="differentnode"))
+    stack.mgr.client.delete(ComposabilityRequest, "w1")
+
+
+def test_async_fabric_end_to_end(stack):
+    """CM-style asynchronous fabric: attach lands after a delay; the
+    controller polls at fabric_wait granularity, not 30 s."""
+    stack.fabric.config = MockFabricConfig(asynchronous=True, attach_latency=0.3)
+    timing = attach_detach_cycle(stack, "async-r", size=1, timeout=30)
+    assert 0.3 * 1000 <= timing["attach_ms"] < 3000
+
+
+def test_node_deletion_garbage_collects_everything(stack):
+    from cro_amd.api.v1alpha1.types import Node
+
+    stack.mgr.client.create(make_request("gc", size=2, target_node="node0"))
+    assert stack.mgr.wait_for(lambda: running(stack, "gc"), timeout=10)
+    stack.mgr.client.delete(Node, "node0")
+    # trigger reconciles via a spec touch (node deletion events are not a
+    # watch source for the request controller — parity with the reference,
+    # which relies on the next reconcile to GC)
+    req = stack.mgr.client.get(ComposabilityRequest, "gc")
+    req.spec.resource.size = 3
+    stack.mgr.client.update(req)
+    assert stack.mgr.wait_for(
+        lambda: stack.mgr.client.try_get(ComposabilityRequest, "gc") is None, timeout=15
+    )
+    assert stack.mgr.wait_for(
+        lambda: stack.mgr.client.list(ComposableResource) == [], timeout=15
+    )

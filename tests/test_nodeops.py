@@ -1,0 +1,225 @@
+"""Node-ops unit tests: KFD topology parsing (fixture sysfs tree), compute
+process scan, CDI spec golden output, capacity checks, debouncer, taints."""
+
+import json
+
+import pytest
+
+from cro_amd.api.v1alpha1.types import DeviceTaintRule, NodeSpecRequirements
+from cro_amd.nodeops.amdgpu import AmdNodeOps, DriverMissing
+from cro_amd.nodeops.cdi_spec import CDISpecWriter
+from cro_amd.nodeops.execs import ExecError, MockNodeExec
+from cro_amd.nodeops.kfd import (
+    KFD_NODES,
+    KFD_PROC,
+    canonical_device_id,
+    enumerate_gpus,
+    gpu_compute_pids,
+)
+from cro_amd.nodeops.nodes import Debouncer, check_node_capacity_sufficient
+from cro_amd.nodeops import taints
+from tests.conftest import make_node, make_resource
+
+NODE = "node0"
+
+
+def kfd_fixture(execer: MockNodeExec, n_gpus: int = 2, node: str = NODE):
+    """Builds a faithful KFD topology tree: one CPU node + n GPU nodes with
+    unique_id, render minors, xGMI links between the GPUs."""
+    execer.set_file(node, f"{KFD_NODES}/0/properties", "cpu_cores_count 96\nsimd_count 0\n")
+    ids = []
+    for i in range(n_gpus):
+        base = f"{KFD_NODES}/{i + 1}"
+        uid = 0xABC0000 + i
+        loc = ((3 + i) << 8)  # bus 03, 04, ...
+        execer.set_file(
+            node,
+            f"{base}/properties",
+            f"simd_count 1024\nunique_id {uid}\ndrm_render_minor {128 + i}\n"
+            f"vendor_id 4098\ndevice_id 29857\nlocation_id {loc}\ndomain 0\n"
+            "gfx_target_version 90500\n",
+        )
+        execer.set_file(node, f"{base}/gpu_id", str(1000 + i))
+        execer.set_file(
+            node, f"{base}/mem_banks/0/properties",
+            "heap_type 1\nsize_in_bytes 309237645312\n",
+        )
+        for j in range(n_gpus):
+            if j != i:
+                execer.set_file(
+                    node, f"{base}/io_links/{j}/properties",
+                    f"type 11\nnode_to {j + 1}\n",
+                )
+        ids.append(canonical_device_id(uid, f"0000:{3 + i:02x}:00.0"))
+    execer.set_file(node, "/sys/module/amdgpu/version", "6.x")
+    return ids
+
+
+def test_enumerate_gpus_from_kfd():
+    ex = MockNodeExec()
+    ids = kfd_fixture(ex, 2)
+    gpus = enumerate_gpus(ex, NODE)
+    assert len(gpus) == 2
+    g0 = gpus[0]
+    assert g0.device_id == ids[0]
+    assert g0.device_id == f"GPU-{0xABC0000:016x}"
+    assert g0.render_minor == 128
+    assert g0.render_path == "/dev/dri/renderD128"
+    assert g0.card_path == "/dev/dri/card0"
+    assert g0.pci_bdf == "0000:03:00.0"
+    assert g0.vram_bytes == 309237645312  # 288 GB HBM3E
+    assert g0.xgmi_peers == [2]
+    assert g0.gpu_id == 1000
+
+
+def test_enumerate_skips_cpu_nodes():
+    ex = MockNodeExec()
+    kfd_fixture(ex, 1)
+    gpus = enumerate_gpus(ex, NODE)
+    assert len(gpus) == 1  # node 0 (CPU) skipped
+
+
+def test_enumerate_without_kfd_raises():
+    ex = MockNodeExec()
+    with pytest.raises(ExecError):
+        enumerate_gpus(ex, NODE)
+
+
+def test_pci_fallback_device_id():
+    assert canonical_device_id(0, "0000:03:00.0") == "GPU-pci-0000:03:00.0"
+
+
+def test_gpu_compute_pids_per_device():
+    ex = MockNodeExec()
+    ex.set_file(NODE, f"{KFD_PROC}/1234/vram_1000", "1048576")
+    ex.set_file(NODE, f"{KFD_PROC}/1234/vram_1001", "0")
+    ex.set_file(NODE, f"{KFD_PROC}/5678/vram_1000", "0")
+    ex.set_file(NODE, f"{KFD_PROC}/5678/vram_1001", "4096")
+    assert gpu_compute_pids(ex, NODE, 1000) == [1234]
+    assert gpu_compute_pids(ex, NODE, 1001) == [5678]
+    assert sorted(gpu_compute_pids(ex, NODE)) == [1234, 5678]
+
+
+def test_gpu_compute_pids_empty_when_no_proc():
+    ex = MockNodeExec()
+    assert gpu_compute_pids(ex, NODE) == []
+
+
+def test_cdi_writer_add_remove(tmp_path):
+    ex = MockNodeExec()
+    kfd_fixture(ex, 2)
+    gpus = enumerate_gpus(ex, NODE)
+    writer = CDISpecWriter(ex, cdi_dir="/etc/cdi")
+    cdi_id = writer.add_device(NODE, gpus[0])
+    assert cdi_id == f"amd.com/gpu={gpus[0].device_id}"
+    spec = json.loads(ex.files[(NODE, "/etc/cdi/amd.com-gpu-cro.json")])
+    assert spec["cdiVersion"] == "0.6.0"
+    assert spec["kind"] == "amd.com/gpu"
+    assert {"path": "/dev/kfd"} in spec["containerEdits"]["deviceNodes"]
+    dev = spec["devices"][0]
+    assert dev["name"] == gpus[0].device_id
+    paths = [d["path"] for d in dev["containerEdits"]["deviceNodes"]]
+    assert "/dev/dri/renderD128" in paths and "/dev/dri/card0" in paths
+    assert dev["annotations"]["cro.amd.com/xgmi-peers"] == "2"
+    assert dev["annotations"]["cro.amd.com/vram-bytes"] == "309237645312"
+    # add second, remove first
+    writer.add_device(NODE, gpus[1])
+    writer.remove_device(NODE, gpus[0].device_id)
+    assert writer.devices(NODE) == [gpus[1].device_id]
+
+
+def test_cdi_add_is_idempotent():
+    ex = MockNodeExec()
+    kfd_fixture(ex, 1)
+    gpus = enumerate_gpus(ex, NODE)
+    writer = CDISpecWriter(ex)
+    writer.add_device(NODE, gpus[0])
+    writer.add_device(NODE, gpus[0])
+    assert len(writer.devices(NODE)) == 1
+
+
+def test_amd_node_ops_driver_and_visibility():
+    ex = MockNodeExec()
+    ids = kfd_fixture(ex, 2)
+    ops = AmdNodeOps(ex, cdi_dir="/etc/cdi")
+    ops.ensure_driver(NODE)
+    assert ops.is_visible(NODE, ids[0])
+    assert not ops.is_visible(NODE, "GPU-nope")
+    ex2 = MockNodeExec()  # no amdgpu module
+    ops2 = AmdNodeOps(ex2)
+    with pytest.raises(DriverMissing):
+        ops2.ensure_driver(NODE)
+
+
+def test_amd_node_ops_simulated_lifecycle():
+    """Non-destructive mode: drain hides the device, compose+rescan restores
+    it — the bench-box lifecycle."""
+    ex = MockNodeExec()
+    ids = kfd_fixture(ex, 2)
+    ops = AmdNodeOps(ex, destructive=False, initially_detached=[ids[0]])
+    assert not ops.is_visible(NODE, ids[0])
+    assert ops.is_visible(NODE, ids[1])
+    ops.simulate_compose(NODE, ids[0])
+    ops.refresh_after_attach(NODE)
+    assert ops.is_visible(NODE, ids[0])
+    ops.drain(NODE, ids[0])
+    assert not ops.is_visible(NODE, ids[0])
+
+
+def test_amd_node_ops_destructive_drain_writes_sysfs():
+    ex = MockNodeExec()
+    ids = kfd_fixture(ex, 2)
+    ops = AmdNodeOps(ex, destructive=True)
+    ops.drain(NODE, ids[0])
+    assert ex.files[(NODE, "/sys/bus/pci/devices/0000:03:00.0/remove")] == "1"
+
+
+def test_amd_node_ops_last_gpu_drain_unloads_module():
+    ex = MockNodeExec()
+    ids = kfd_fixture(ex, 1)
+    ex.set_command(("modprobe", "-r", "amdgpu"), (0, "", ""))
+    ops = AmdNodeOps(ex, destructive=True)
+    ops.drain(NODE, ids[0])
+    assert ("run", NODE, ("modprobe", "-r", "amdgpu")) in ex.calls
+    assert ex.files[(NODE, "/sys/bus/pci/devices/0000:03:00.0/remove")] == "1"
+
+
+def test_capacity_check(client):
+    make_node(client, "node0", milli_cpu=8000, memory=1 << 30, pods=10)
+    assert check_node_capacity_sufficient(
+        client, "node0", NodeSpecRequirements(milli_cpu=4000)
+    )
+    assert not check_node_capacity_sufficient(
+        client, "node0", NodeSpecRequirements(milli_cpu=16000)
+    )
+    assert not check_node_capacity_sufficient(
+        client, "node0", NodeSpecRequirements(allowed_pod_number=100)
+    )
+
+
+def test_debouncer():
+    import time
+
+    d = Debouncer(interval=0.1)
+    calls = []
+    assert d("k", lambda: calls.append(1)) is True
+    assert d("k", lambda: calls.append(2)) is False  # debounced
+    assert d("other", lambda: calls.append(3)) is True  # separate key
+    time.sleep(0.11)
+    assert d("k", lambda: calls.append(4)) is True
+    assert calls == [1, 3, 4]
+
+
+def test_taint_helpers(client):
+    r = make_resource("gpu-1")
+    r.status.device_id = "GPU-x"
+    taints.create_device_taint(client, r)
+    taints.create_device_taint(client, r)  # idempotent
+    assert taints.has_device_taint(client, r)
+    rules = client.list(DeviceTaintRule)
+    assert len(rules) == 1
+    assert rules[0].spec.device_uuid == "GPU-x"
+    assert rules[0].spec.effect == "NoSchedule"
+    taints.delete_device_taint(client, r)
+    taints.delete_device_taint(client, r)  # idempotent
+    assert not taints.has_device_taint(client, r)

@@ -1,0 +1,68 @@
+"""Admission validation: the three webhook rules
+(composabilityrequest_webhook.go:84-131 parity), exercised through the store
+admission chain the way the reference exercises the real TLS endpoint via
+envtest WebhookInstallOptions."""
+
+import pytest
+
+from cro_amd.api.v1alpha1.types import ComposabilityRequest, ScalarResourceStatus
+from cro_amd.runtime.errors import AdmissionDenied
+from cro_amd.webhook.validator import admission_validator
+from tests.conftest import make_request
+
+
+@pytest.fixture
+def guarded_client(client, store):
+    store.register_admission("ComposabilityRequest", admission_validator(client))
+    return client
+
+
+def test_differentnode_with_target_node_rejected(guarded_client):
+    req = make_request("r1", policy="differentnode", target_node="node0")
+    with pytest.raises(AdmissionDenied, match="TargetNode cannot be specified"):
+        guarded_client.create(req)
+
+
+def test_duplicate_differentnode_type_model_rejected(guarded_client):
+    guarded_client.create(make_request("r1", policy="differentnode"))
+    with pytest.raises(AdmissionDenied, match="already exists"):
+        guarded_client.create(make_request("r2", policy="differentnode"))
+
+
+def test_differentnode_different_model_allowed(guarded_client):
+    guarded_client.create(make_request("r1", policy="differentnode", model="mi355x"))
+    guarded_client.create(make_request("r2", policy="differentnode", model="mi300x"))
+
+
+def test_duplicate_samenode_target_type_model_rejected(guarded_client):
+    guarded_client.create(make_request("r1", target_node="node0"))
+    with pytest.raises(AdmissionDenied, match="already exists"):
+        guarded_client.create(make_request("r2", target_node="node0"))
+
+
+def test_samenode_different_nodes_allowed(guarded_client):
+    guarded_client.create(make_request("r1", target_node="node0"))
+    guarded_client.create(make_request("r2", target_node="node1"))
+
+
+def test_samenode_implicit_target_resolved_from_status(guarded_client):
+    # r1 has no explicit target but its status pins node0
+    r1 = guarded_client.create(make_request("r1"))
+    r1.status.resources = {"gpu-x": ScalarResourceStatus(node_name="node0")}
+    guarded_client.update_status(r1)
+    with pytest.raises(AdmissionDenied, match="already exists"):
+        guarded_client.create(make_request("r2", target_node="node0"))
+
+
+def test_update_validated_too(guarded_client):
+    guarded_client.create(make_request("r1", target_node="node0"))
+    r2 = guarded_client.create(make_request("r2", target_node="node1"))
+    r2.spec.resource.target_node = "node0"
+    with pytest.raises(AdmissionDenied, match="already exists"):
+        guarded_client.update(r2)
+
+
+def test_self_update_not_a_conflict(guarded_client):
+    created = guarded_client.create(make_request("r1", target_node="node0"))
+    created.spec.resource.size = 4
+    guarded_client.update(created)  # same name — excluded from conflict scan
