@@ -90,7 +90,7 @@ def enumerate_gpus(execer: NodeExec, node: str) -> List[GPUDevice]:
     """Enumerate GPUs from KFD topology; raises ExecError if KFD is absent."""
     try:
         entries = execer.list_dir(node, KFD_NODES)
-    except FileNotFoundError:
+    except (FileNotFoundError, PermissionError, OSError):
         raise ExecError("KFD topology not present (amdgpu driver not loaded?)")
 
     gpus: List[GPUDevice] = []
@@ -98,13 +98,15 @@ def enumerate_gpus(execer: NodeExec, node: str) -> List[GPUDevice]:
         base = f"{KFD_NODES}/{entry}"
         try:
             props = _parse_properties(execer.read_file(node, f"{base}/properties"))
-        except FileNotFoundError:
+        except (FileNotFoundError, PermissionError, OSError):
+            # unassigned GPUs are cgroup-hidden: their node dirs exist but
+            # reads fail with EPERM — they are not ours to see
             continue
         if props.get("simd_count", 0) <= 0:
             continue  # CPU node
         try:
             gpu_id = int(execer.read_file(node, f"{base}/gpu_id").strip())
-        except (FileNotFoundError, ValueError):
+        except (FileNotFoundError, PermissionError, OSError, ValueError):
             gpu_id = 0
         unique_id = props.get("unique_id", 0)
         bdf = _pci_bdf(props)
@@ -127,7 +129,7 @@ def enumerate_gpus(execer: NodeExec, node: str) -> List[GPUDevice]:
                 )
                 if bprops.get("heap_type", 0) in (1, 2):
                     dev.vram_bytes += bprops.get("size_in_bytes", 0)
-        except FileNotFoundError:
+        except (FileNotFoundError, PermissionError, OSError):
             pass
         # xGMI peer links
         try:
@@ -137,7 +139,7 @@ def enumerate_gpus(execer: NodeExec, node: str) -> List[GPUDevice]:
                 )
                 if lprops.get("type", 0) == IOLINK_TYPE_XGMI:
                     dev.xgmi_peers.append(lprops.get("node_to", -1))
-        except FileNotFoundError:
+        except (FileNotFoundError, PermissionError, OSError):
             pass
         gpus.append(dev)
     return gpus
@@ -182,7 +184,7 @@ def gpu_compute_pids(execer: NodeExec, node: str, gpu_id: Optional[int] = None) 
     """
     try:
         pids = execer.list_dir(node, KFD_PROC)
-    except FileNotFoundError:
+    except (FileNotFoundError, PermissionError, OSError):
         return []
     result: List[int] = []
     for pid in pids:
@@ -195,7 +197,7 @@ def gpu_compute_pids(execer: NodeExec, node: str, gpu_id: Optional[int] = None) 
             vram = execer.read_file(node, f"{KFD_PROC}/{pid}/vram_{gpu_id}").strip()
             if int(vram) > 0:
                 result.append(int(pid))
-        except (FileNotFoundError, ValueError):
+        except (FileNotFoundError, PermissionError, OSError, ValueError):
             # no per-GPU attribution available → count it conservatively
             result.append(int(pid))
     return result
