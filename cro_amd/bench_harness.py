@@ -191,7 +191,19 @@ def attach_detach_cycle(stack: LocalStack, name: str, size: int = 1, timeout: fl
     )
     t2 = time.monotonic()
     if not ok:
-        raise RuntimeError(f"request {name} did not tear down in {timeout}s")
+        from .api.v1alpha1.types import ComposableResource
+
+        children = [
+            (c.metadata.name, c.status.state, c.status.error)
+            for c in mgr.client.list(
+                ComposableResource, {"app.kubernetes.io/managed-by": name}
+            )
+        ]
+        cur = mgr.client.try_get(ComposabilityRequest, name)
+        raise RuntimeError(
+            f"request {name} did not tear down in {timeout}s "
+            f"(state={cur.status.state if cur else 'gone'}, children={children})"
+        )
     return {"attach_ms": (t1 - t0) * 1e3, "detach_ms": (t2 - t1) * 1e3}
 
 

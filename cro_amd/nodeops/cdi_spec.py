@@ -60,13 +60,13 @@ class CDISpecWriter:
         """Add (or refresh) one composed GPU; returns its CDI device id."""
         with self._lock:
             spec = self._load(node)
+            device_nodes = [{"path": gpu.render_path}]
+            if gpu.card_path is not None:
+                device_nodes.append({"path": gpu.card_path})
             entry = {
                 "name": gpu.device_id,
                 "containerEdits": {
-                    "deviceNodes": [
-                        {"path": gpu.render_path},
-                        {"path": gpu.card_path},
-                    ],
+                    "deviceNodes": device_nodes,
                 },
                 "annotations": {
                     "cro.amd.com/pci-bdf": gpu.pci_bdf,

@@ -47,15 +47,20 @@ class GPUDevice:
     vram_bytes: int = 0
     gfx_target: str = ""
     xgmi_peers: List[int] = field(default_factory=list)  # peer kfd node ids
+    card_index: Optional[int] = None  # resolved from /sys/class/drm by PCI
 
     @property
     def render_path(self) -> str:
         return f"/dev/dri/renderD{self.render_minor}"
 
     @property
-    def card_path(self) -> str:
-        # card minor = render minor - 128 by DRM convention
-        return f"/dev/dri/card{self.render_minor - 128}"
+    def card_path(self) -> Optional[str]:
+        # card numbering does NOT track render minors (a node exposing one
+        # GPU of many can have renderD144 + card16): only a PCI match on
+        # /sys/class/drm/card*/device is authoritative
+        if self.card_index is None:
+            return None
+        return f"/dev/dri/card{self.card_index}"
 
 
 def canonical_device_id(unique_id: int, pci_bdf: str) -> str:
@@ -86,12 +91,33 @@ def _pci_bdf(props: Dict[str, int]) -> str:
     return f"{domain:04x}:{bus:02x}:{dev:02x}.{fn:x}"
 
 
+def _drm_card_by_bdf(execer: NodeExec, node: str) -> Dict[str, int]:
+    """Map PCI DBDF → /dev/dri/card index via /sys/class/drm/card*/device."""
+    mapping: Dict[str, int] = {}
+    try:
+        entries = execer.list_dir(node, "/sys/class/drm")
+    except (FileNotFoundError, PermissionError, OSError):
+        return mapping
+    for entry in entries:
+        if not (entry.startswith("card") and entry[4:].isdigit()):
+            continue
+        try:
+            uevent = execer.read_file(node, f"/sys/class/drm/{entry}/device/uevent")
+        except (FileNotFoundError, PermissionError, OSError):
+            continue
+        for line in uevent.splitlines():
+            if line.startswith("PCI_SLOT_NAME="):
+                mapping[line.split("=", 1)[1].strip().lower()] = int(entry[4:])
+    return mapping
+
+
 def enumerate_gpus(execer: NodeExec, node: str) -> List[GPUDevice]:
     """Enumerate GPUs from KFD topology; raises ExecError if KFD is absent."""
     try:
         entries = execer.list_dir(node, KFD_NODES)
     except (FileNotFoundError, PermissionError, OSError):
         raise ExecError("KFD topology not present (amdgpu driver not loaded?)")
+    card_map = _drm_card_by_bdf(execer, node)
 
     gpus: List[GPUDevice] = []
     for entry in entries:
@@ -120,6 +146,7 @@ def enumerate_gpus(execer: NodeExec, node: str) -> List[GPUDevice]:
             vendor_id=props.get("vendor_id", 0),
             pci_device_id=props.get("device_id", 0),
             gfx_target=str(props.get("gfx_target_version", "")),
+            card_index=card_map.get(bdf.lower()),
         )
         # VRAM from mem_banks (heap_type 1/2 = FB public/private)
         try:
@@ -193,11 +220,16 @@ def gpu_compute_pids(execer: NodeExec, node: str, gpu_id: Optional[int] = None) 
         if gpu_id is None:
             result.append(int(pid))
             continue
+        # per-device attribution: a pid loads THIS gpu iff its vram_<gpu_id>
+        # file reads > 0.  A missing/unreadable file means the process has no
+        # context on this device (other tenants' processes on a shared node
+        # are visible in the proc dir but their per-GPU files are not ours to
+        # read) — do NOT count those, or detach wedges forever on shared
+        # machines.  The node-wide check (gpu_id=None) stays conservative.
         try:
             vram = execer.read_file(node, f"{KFD_PROC}/{pid}/vram_{gpu_id}").strip()
             if int(vram) > 0:
                 result.append(int(pid))
         except (FileNotFoundError, PermissionError, OSError, ValueError):
-            # no per-GPU attribution available → count it conservatively
-            result.append(int(pid))
+            continue
     return result

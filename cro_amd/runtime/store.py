@@ -158,6 +158,20 @@ class InMemoryStore:
         with self._lock:
             stored = self._require(obj.kind, obj.metadata.name)
             self._check_rv(stored, obj)
+            # no-op update: no RV bump, no watch event (apiserver parity)
+            incoming_meta = obj.metadata.model_copy(
+                update={
+                    "resourceVersion": stored.metadata.resourceVersion,
+                    "uid": stored.metadata.uid,
+                    "generation": stored.metadata.generation,
+                    "creationTimestamp": stored.metadata.creationTimestamp,
+                    "deletionTimestamp": stored.metadata.deletionTimestamp,
+                }
+            )
+            if incoming_meta == stored.metadata and getattr(obj, "spec", None) == getattr(
+                stored, "spec", None
+            ):
+                return stored.model_copy(deep=True)
             self._admit("UPDATE", stored, obj)
             new = obj
             if hasattr(stored, "status"):
@@ -175,10 +189,18 @@ class InMemoryStore:
             return new.model_copy(deep=True)
 
     def update_status(self, obj: K8sObject) -> K8sObject:
-        """Status-subresource update: only .status is applied."""
+        """Status-subresource update: only .status is applied.
+
+        A no-op write (identical status) returns the stored object without
+        bumping resourceVersion or firing a watch event — apiserver parity;
+        without this, a reconciler that re-writes the same error on every
+        retry generates its own watch events and hot-loops.
+        """
         with self._lock:
             stored = self._require(obj.kind, obj.metadata.name)
             self._check_rv(stored, obj)
+            if stored.status == obj.status:
+                return stored.model_copy(deep=True)
             new = stored.model_copy(deep=True)
             new.status = obj.status.model_copy(deep=True)
             new.metadata.resourceVersion = str(next(self._rv))

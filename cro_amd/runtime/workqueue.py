@@ -67,7 +67,9 @@ class RateLimitedQueue:
         with self._cond:
             n = self._failures.get(key, 0)
             self._failures[key] = n + 1
-        self.add_after(key, min(self._base * (2**n), self._max))
+        # clamp the exponent: the failure count is unbounded and 2**n is an
+        # arbitrary-precision int that overflows float math past ~2**1024
+        self.add_after(key, min(self._base * (2 ** min(n, 32)), self._max))
 
     def forget(self, key: Hashable) -> None:
         with self._cond:

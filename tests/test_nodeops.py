@@ -40,6 +40,12 @@ def kfd_fixture(execer: MockNodeExec, n_gpus: int = 2, node: str = NODE):
             "gfx_target_version 90500\n",
         )
         execer.set_file(node, f"{base}/gpu_id", str(1000 + i))
+        # DRM card mapping is by PCI slot, deliberately NOT render-128
+        # (real boxes renumber: renderD144 ↔ card16)
+        execer.set_file(
+            node, f"/sys/class/drm/card{i}/device/uevent",
+            f"DRIVER=amdgpu\nPCI_SLOT_NAME=0000:{3 + i:02x}:00.0\n",
+        )
         execer.set_file(
             node, f"{base}/mem_banks/0/properties",
             "heap_type 1\nsize_in_bytes 309237645312\n",
