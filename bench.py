@@ -41,6 +41,13 @@ def main() -> int:
     p.add_argument("--size", type=int, default=1, help="devices per request")
     p.add_argument("--mode", default="DRA", choices=["DRA", "DEVICE_PLUGIN"])
     p.add_argument("--no-probe", action="store_true")
+    p.add_argument(
+        "--force-detach",
+        action="store_true",
+        help="set force_detach on the CR spec (skips the detach load check; "
+        "for test environments where another process legitimately holds a "
+        "KFD context on the bench GPU)",
+    )
     args = p.parse_args()
 
     import torch
@@ -88,7 +95,7 @@ def main() -> int:
 
     # -- warmup (includes HIP context + probe first-touch) ------------------
     for i in range(args.warmup):
-        attach_detach_cycle(stack, f"warm-{rank}-{i}", size=args.size)
+        attach_detach_cycle(stack, f"warm-{rank}-{i}", size=args.size, force_detach=args.force_detach)
 
     barrier()
     rec0 = reconcile_count(stack)
@@ -96,7 +103,11 @@ def main() -> int:
 
     samples = []
     for i in range(args.steps):
-        samples.append(attach_detach_cycle(stack, f"step-{rank}-{i}", size=args.size))
+        samples.append(
+            attach_detach_cycle(
+                stack, f"step-{rank}-{i}", size=args.size, force_detach=args.force_detach
+            )
+        )
 
     barrier()
     t_end = time.monotonic()

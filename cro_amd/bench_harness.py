@@ -153,7 +153,7 @@ def build_local_stack(
     )
 
 
-def attach_detach_cycle(stack: LocalStack, name: str, size: int = 1, timeout: float = 60.0) -> dict:
+def attach_detach_cycle(stack: LocalStack, name: str, size: int = 1, timeout: float = 60.0, force_detach: bool = False) -> dict:
     """One full ComposabilityRequest lifecycle; returns timing samples.
 
     attach_ms = create → Running (every device Online with CDI written);
@@ -163,7 +163,8 @@ def attach_detach_cycle(stack: LocalStack, name: str, size: int = 1, timeout: fl
     req = ComposabilityRequest(
         spec=ComposabilityRequestSpec(
             resource=ScalarResourceDetails(
-                type="gpu", model="mi355x", size=size, target_node=stack.node_name
+                type="gpu", model="mi355x", size=size, target_node=stack.node_name,
+                force_detach=force_detach,
             )
         )
     )

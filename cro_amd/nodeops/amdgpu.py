@@ -162,12 +162,18 @@ class AmdNodeOps(NodeOps):
         pids = gpu_compute_pids(self.execer, node, gpu_id)
         # node-agent self-exemption: when operating on the local node, our own
         # process holds a KFD context for the health probe — it must not block
-        # the detach it is itself orchestrating
-        import os as _os
+        # the detach it is itself orchestrating.  KFD keys its proc dir by
+        # HOST pid (a containerized agent sees namespaced pids), hence
+        # self_host_pid() rather than os.getpid().
         from .execs import LocalNodeExec as _Local
+        from .kfd import self_host_pid
 
         if isinstance(self.execer, _Local):
-            pids = [p for p in pids if p != _os.getpid()]
+            own = {self_host_pid()}
+            import os as _os
+
+            own.add(_os.getpid())
+            pids = [p for p in pids if p not in own]
         if pids:
             scope = f"device {device_id}" if device_id else f"node {node}"
             raise GPULoadsPresent(f"{scope} has active KFD compute processes: {pids}")

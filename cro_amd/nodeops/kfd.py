@@ -202,6 +202,30 @@ def enumerate_gpus_amdsmi(execer: NodeExec, node: str) -> List[GPUDevice]:
     return gpus
 
 
+def self_host_pid() -> int:
+    """This process's pid in the HOST pid namespace.
+
+    /sys/class/kfd/kfd/proc is keyed by host pids; a containerized node
+    agent sees namespaced pids from os.getpid(), so self-exemption in the
+    load check needs the host pid.  The first line of /proc/self/sched leaks
+    it as ``comm (HOSTPID, #threads: N)``; CRO_SELF_KFD_PID overrides.
+    """
+    import os
+    import re
+
+    env = os.environ.get("CRO_SELF_KFD_PID", "")
+    if env.isdigit():
+        return int(env)
+    try:
+        with open("/proc/self/sched") as f:
+            m = re.search(r"\((\d+),", f.readline())
+        if m:
+            return int(m.group(1))
+    except OSError:
+        pass
+    return os.getpid()
+
+
 def gpu_compute_pids(execer: NodeExec, node: str, gpu_id: Optional[int] = None) -> List[int]:
     """PIDs with open KFD compute contexts (the amdgpu-native analog of
     ``nvidia-smi --query-compute-apps``, gpus.go:241-350).

@@ -128,7 +128,10 @@ def test_cdi_spec_contents_on_real_gpu():
 def test_bench_one_gpu_quick():
     _require_gpu()
     proc = subprocess.run(
-        [sys.executable, "bench.py", "--steps", "5", "--warmup", "2"],
+        # --force-detach: this pytest process already holds a KFD context
+        # on the bench GPU (probe tests above), which the subprocess's
+        # detach load-check would legitimately see as foreign load
+        [sys.executable, "bench.py", "--steps", "5", "--warmup", "2", "--force-detach"],
         cwd=REPO,
         capture_output=True,
         text=True,
