@@ -260,6 +260,21 @@ extern "C" int cro_probe_mfma_f32(int device, const float* A, const float* B,
   return 0;
 }
 
+// Raw VRAM alloc/free for the self-identification fingerprint
+// (cro_amd/nodeops/kfd.py resolve_self_kfd_pid): the operator finds its own
+// host pid in /sys/class/kfd/kfd/proc by watching which vram_<gpu_id> file
+// grows by a marker-sized hipMalloc.
+extern "C" void* cro_probe_alloc(int device, long long bytes) {
+  if (hipSetDevice(device) != hipSuccess) return nullptr;
+  void* p = nullptr;
+  if (hipMalloc(&p, (size_t)bytes) != hipSuccess) return nullptr;
+  if (hipMemset(p, 1, (size_t)bytes) != hipSuccess) { (void)hipFree(p); return nullptr; }
+  if (hipDeviceSynchronize() != hipSuccess) { (void)hipFree(p); return nullptr; }
+  return p;
+}
+
+extern "C" void cro_probe_free(void* p) { (void)hipFree(p); }
+
 extern "C" int cro_probe_device_count(void) {
   int count = 0;
   if (hipGetDeviceCount(&count) != hipSuccess) return -1;

@@ -106,6 +106,8 @@ class AmdNodeOps(NodeOps):
         # daemonset-restart analogs, keyed by component name
         self.restart_hooks = restart_hooks or {}
         self.probe_fn = probe_fn  # probe_fn(gpu: GPUDevice) -> result dict
+        # gpu_id -> our own host pid holding a KFD context there (or None)
+        self._self_pid_cache: Dict[int, Optional[int]] = {}
 
     # -- driver ------------------------------------------------------------
 
@@ -160,23 +162,51 @@ class AmdNodeOps(NodeOps):
                 return  # device already gone — nothing can be loading it
             gpu_id = gpu.gpu_id
         pids = gpu_compute_pids(self.execer, node, gpu_id)
-        # node-agent self-exemption: when operating on the local node, our own
-        # process holds a KFD context for the health probe — it must not block
-        # the detach it is itself orchestrating.  KFD keys its proc dir by
-        # HOST pid (a containerized agent sees namespaced pids), hence
-        # self_host_pid() rather than os.getpid().
-        from .execs import LocalNodeExec as _Local
-        from .kfd import self_host_pid
-
-        if isinstance(self.execer, _Local):
-            own = {self_host_pid()}
-            import os as _os
-
-            own.add(_os.getpid())
-            pids = [p for p in pids if p not in own]
+        pids = [p for p in pids if p not in self._own_pids(node, gpu_id)]
         if pids:
             scope = f"device {device_id}" if device_id else f"node {node}"
             raise GPULoadsPresent(f"{scope} has active KFD compute processes: {pids}")
+
+    def _own_pids(self, node: str, gpu_id: Optional[int]) -> Set[int]:
+        """Node-agent self-exemption for the load check: the agent's own
+        process holds a KFD context (health probe / torch) and must not
+        block the detach it is itself orchestrating.
+
+        /sys/class/kfd/kfd/proc is keyed by HOST pids; a containerized
+        agent's os.getpid() is namespaced, so the authoritative mapping is
+        the VRAM-fingerprint resolver (kfd.resolve_self_kfd_pid), cached per
+        device.  Applies only when operating on the local node.
+        """
+        from .execs import LocalNodeExec as _Local
+        from .kfd import resolve_self_kfd_pid, self_host_pid
+
+        if not isinstance(self.execer, _Local):
+            return set()
+        import os as _os
+
+        own: Set[int] = {_os.getpid(), self_host_pid()}
+        if gpu_id is not None:
+            cached = self._self_pid_cache.get(gpu_id)
+            if cached is None and gpu_id not in self._self_pid_cache:
+                hip_dev = None
+                try:
+                    from .probe import hip_device_for_bdf, load_library
+
+                    if load_library(required=False) is not None:
+                        for g in self.enumerate(node):
+                            if g.gpu_id == gpu_id:
+                                hip_dev = hip_device_for_bdf(g.pci_bdf)
+                                break
+                        if hip_dev is not None:
+                            cached = resolve_self_kfd_pid(
+                                self.execer, node, gpu_id, hip_dev
+                            )
+                except Exception as exc:
+                    log.debug("self-pid resolution failed: %s", exc)
+                self._self_pid_cache[gpu_id] = cached
+            if cached is not None:
+                own.add(cached)
+        return own
 
     # -- drain / attach refresh -------------------------------------------
 

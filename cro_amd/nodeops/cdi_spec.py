@@ -61,7 +61,9 @@ class CDISpecWriter:
         with self._lock:
             spec = self._load(node)
             device_nodes = [{"path": gpu.render_path}]
-            if gpu.card_path is not None:
+            # the card node is optional (compute needs kfd+render only) and a
+            # containerized agent may not have it mapped — emit only if real
+            if gpu.card_path is not None and self.execer.path_exists(node, gpu.card_path):
                 device_nodes.append({"path": gpu.card_path})
             entry = {
                 "name": gpu.device_id,

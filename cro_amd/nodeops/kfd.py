@@ -203,12 +203,13 @@ def enumerate_gpus_amdsmi(execer: NodeExec, node: str) -> List[GPUDevice]:
 
 
 def self_host_pid() -> int:
-    """This process's pid in the HOST pid namespace.
+    """Best-effort host-namespace pid of this process (see
+    :func:`resolve_self_kfd_pid` for the authoritative method).
 
-    /sys/class/kfd/kfd/proc is keyed by host pids; a containerized node
-    agent sees namespaced pids from os.getpid(), so self-exemption in the
-    load check needs the host pid.  The first line of /proc/self/sched leaks
-    it as ``comm (HOSTPID, #threads: N)``; CRO_SELF_KFD_PID overrides.
+    /sys/class/kfd/kfd/proc is keyed by HOST pids; a containerized node
+    agent sees namespaced pids from os.getpid().  CRO_SELF_KFD_PID
+    overrides; /proc/self/sched leaks the host pid on some kernels (modern
+    ones show the namespaced pid — then this falls back to os.getpid()).
     """
     import os
     import re
@@ -224,6 +225,66 @@ def self_host_pid() -> int:
     except OSError:
         pass
     return os.getpid()
+
+
+def _read_vram_map(execer: NodeExec, node: str, gpu_id: int) -> Dict[int, int]:
+    out: Dict[int, int] = {}
+    try:
+        pids = execer.list_dir(node, KFD_PROC)
+    except (FileNotFoundError, PermissionError, OSError):
+        return out
+    for pid in pids:
+        if not pid.isdigit():
+            continue
+        try:
+            out[int(pid)] = int(
+                execer.read_file(node, f"{KFD_PROC}/{pid}/vram_{gpu_id}").strip()
+            )
+        except (FileNotFoundError, PermissionError, OSError, ValueError):
+            pass
+    return out
+
+
+def resolve_self_kfd_pid(
+    execer: NodeExec, node: str, gpu_id: int, hip_device: int
+) -> Optional[int]:
+    """Authoritative host-pid self-identification via a VRAM fingerprint.
+
+    Allocates a marker-sized VRAM buffer on the device and returns the pid
+    of the single /sys/class/kfd/kfd/proc entry whose ``vram_<gpu_id>``
+    grew by at least the marker — that entry is this process (cgroup device
+    isolation guarantees no other container can touch our GPU, and no other
+    process of ours allocates concurrently).  Retries with distinct marker
+    sizes to reject coincidental growth.  Returns None when no probe
+    library / GPU is available.
+    """
+    import os
+
+    try:
+        from .probe import vram_alloc, vram_free
+    except Exception:
+        return None
+
+    for attempt in range(3):
+        marker = (48 << 20) + ((os.getpid() + attempt * 7919) % 4096) * 8192
+        before = _read_vram_map(execer, node, gpu_id)
+        handle = None
+        try:
+            handle = vram_alloc(hip_device, marker)
+            if not handle:
+                return None
+            after = _read_vram_map(execer, node, gpu_id)
+        finally:
+            if handle:
+                vram_free(handle)
+        grown = [
+            pid
+            for pid, v in after.items()
+            if v - before.get(pid, 0) >= int(marker * 0.9)
+        ]
+        if len(grown) == 1:
+            return grown[0]
+    return None
 
 
 def gpu_compute_pids(execer: NodeExec, node: str, gpu_id: Optional[int] = None) -> List[int]:

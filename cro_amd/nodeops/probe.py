@@ -55,8 +55,24 @@ def load_library(required: Optional[bool] = None) -> Optional[ctypes.CDLL]:
     lib.cro_probe_device_count.restype = ctypes.c_int
     lib.cro_probe_pci_bus_id.argtypes = [ctypes.c_int, ctypes.c_char_p, ctypes.c_int]
     lib.cro_probe_pci_bus_id.restype = ctypes.c_int
+    lib.cro_probe_alloc.argtypes = [ctypes.c_int, ctypes.c_longlong]
+    lib.cro_probe_alloc.restype = ctypes.c_void_p
+    lib.cro_probe_free.argtypes = [ctypes.c_void_p]
+    lib.cro_probe_free.restype = None
     _lib = lib
     return lib
+
+
+def vram_alloc(device: int, nbytes: int):
+    """Allocate+touch VRAM on a device; returns an opaque handle (None on
+    failure).  Used by the self-pid fingerprint."""
+    lib = load_library(required=True)
+    return lib.cro_probe_alloc(device, nbytes)
+
+
+def vram_free(handle) -> None:
+    if handle:
+        load_library(required=True).cro_probe_free(handle)
 
 
 def device_count() -> int:
