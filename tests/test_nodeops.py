@@ -46,6 +46,7 @@ def kfd_fixture(execer: MockNodeExec, n_gpus: int = 2, node: str = NODE):
             node, f"/sys/class/drm/card{i}/device/uevent",
             f"DRIVER=amdgpu\nPCI_SLOT_NAME=0000:{3 + i:02x}:00.0\n",
         )
+        execer.set_file(node, f"/dev/dri/card{i}", "")  # device file present
         execer.set_file(
             node, f"{base}/mem_banks/0/properties",
             "heap_type 1\nsize_in_bytes 309237645312\n",
