@@ -24,13 +24,27 @@ from .upstreamsyncer import UpstreamSyncer
 
 
 def _status_changed(ev: WatchEvent) -> bool:
-    """UPDATE events pass only when .status changed; CREATE/DELETE are
-    dropped (resourceStatusUpdatePredicate parity, :658-678)."""
+    """UPDATE events pass only when .status changed (resourceStatusUpdate-
+    Predicate parity, :658-678); DELETED events of managed children also
+    pass — a deliberate improvement over the reference, whose Cleaning and
+    Updating states poll at 30 s because child deletions never wake the
+    parent (:612)."""
+    if ev.type == "DELETED":
+        return bool(ev.object.metadata.labels.get("app.kubernetes.io/managed-by"))
     if ev.type != "MODIFIED":
         return False
     if ev.old_object is None:
         return True
     return getattr(ev.object, "status", None) != getattr(ev.old_object, "status", None)
+
+
+def _child_event_mapper(ev: WatchEvent):
+    """Child status changes reconcile under the child's name (the dual-kind
+    sync path); child deletions wake the owning request directly."""
+    if ev.type == "DELETED":
+        parent = ev.object.metadata.labels.get("app.kubernetes.io/managed-by", "")
+        return [parent] if parent else []
+    return [ev.object.metadata.name]
 
 
 def build_manager(
@@ -66,7 +80,11 @@ def build_manager(
             request_reconciler,
             sources=[
                 Source(kind="ComposabilityRequest"),
-                Source(kind="ComposableResource", predicate=_status_changed),
+                Source(
+                    kind="ComposableResource",
+                    predicate=_status_changed,
+                    mapper=_child_event_mapper,
+                ),
             ],
             max_concurrent_reconciles=max_concurrent_reconciles,
         )
