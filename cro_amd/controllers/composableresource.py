@@ -203,10 +203,12 @@ class ComposableResourceReconciler(Reconciler):
         if not visible:
             return Result(requeue_after=self.config.attach_visible_wait)
 
-        # device enumerable → emit CDI spec and (optionally) verify compute
-        cdi_id = self.node_ops.write_cdi(node, resource.status.device_id)
-        if cdi_id:
-            resource.status.cdi_device_id = cdi_id
+        # device enumerable → emit the Container Device Interface spec and
+        # (optionally) verify compute.  NOTE: status.cdi_device_id is the
+        # *Composable Disaggregated Infrastructure* id the fabric handed out
+        # (FM detach keys on it, fm/client.go:231-242) — the container-CDI
+        # spec name is amd.com/gpu=<device_id>, never stored over it.
+        self.node_ops.write_cdi(node, resource.status.device_id)
         probe = self.node_ops.health_probe(node, resource.status.device_id)
         if probe is not None and not probe.get("ok", True):
             raise FabricError(f"gfx950 health probe failed: {probe}")
