@@ -1,0 +1,142 @@
+"""Operator process entrypoint (cmd/main.go parity).
+
+Flags mirror the reference (cmd/main.go:68-81) where they still apply, plus
+the standalone-mode additions:
+
+    python -m cro_amd.cmd.main \
+        --metrics-bind-address :8443 --health-probe-bind-address :8081 \
+        --api-bind-address :8080 --leader-elect \
+        --node <name> [--cdi-dir /etc/cdi] [--destructive]
+
+Env config (composableresource_adapter.go:40-76 and backends):
+  DEVICE_RESOURCE_TYPE, CDI_PROVIDER_TYPE, FTI_CDI_API_TYPE,
+  FTI_CDI_ENDPOINT, FTI_CDI_TENANT_ID, FTI_CDI_CLUSTER_ID,
+  NEC_CDIM_IP, LAYOUT_APPLY_PORT, CONFIGURATION_MANAGER_PORT,
+  NEC_PROVISIONAL_GPU_UUID, SUNFISH_ENDPOINT, ENABLE_WEBHOOKS,
+  CRO_FTI_USERNAME/PASSWORD/CLIENT_ID/CLIENT_SECRET/REALM.
+
+Leader election uses an exclusive flock on a lock file — the single-node
+analog of the reference's lease-based election (cmd/main.go:137-155); in a
+multi-replica cluster deployment the lease path belongs to the kube client
+integration.
+"""
+
+from __future__ import annotations
+
+import argparse
+import fcntl
+import logging
+import os
+import signal
+import sys
+import threading
+
+
+def parse_port(addr: str, default: int) -> int:
+    if not addr:
+        return default
+    return int(addr.rsplit(":", 1)[-1])
+
+
+def main(argv=None) -> int:
+    p = argparse.ArgumentParser(prog="cro-amd-operator")
+    p.add_argument("--metrics-bind-address", default=":8443")
+    p.add_argument("--health-probe-bind-address", default=":8081")
+    p.add_argument("--api-bind-address", default=":8080")
+    p.add_argument("--leader-elect", action="store_true")
+    p.add_argument("--leader-elect-lock", default="/var/run/cro-amd/leader.lock")
+    p.add_argument("--node", default=os.environ.get("NODE_NAME", ""),
+                   help="local node name this operator instance manages")
+    p.add_argument("--cdi-dir", default="/etc/cdi")
+    p.add_argument("--destructive", action="store_true",
+                   help="perform real PCI remove/rescan and module unload")
+    p.add_argument("--max-concurrent-reconciles", type=int, default=8)
+    p.add_argument("--syncer-period", type=float, default=60.0)
+    p.add_argument("--zap-log-level", default="info")
+    args = p.parse_args(argv)
+
+    logging.basicConfig(
+        level=getattr(logging, args.zap_log_level.upper(), logging.INFO),
+        format="%(asctime)s %(levelname)s %(name)s %(message)s",
+    )
+    log = logging.getLogger("cro_amd.main")
+
+    lock_file = None
+    if args.leader_elect:
+        os.makedirs(os.path.dirname(args.leader_elect_lock), exist_ok=True)
+        lock_file = open(args.leader_elect_lock, "w")
+        log.info("waiting for leader lock %s", args.leader_elect_lock)
+        fcntl.flock(lock_file, fcntl.LOCK_EX)
+        log.info("acquired leadership")
+
+    from ..controllers import build_manager
+    from ..fabric.adapter import new_adapter
+    from ..nodeops.amdgpu import AmdNodeOps
+    from ..nodeops.execs import LocalNodeExec
+
+    adapter = new_adapter()
+
+    mgr = build_manager(
+        adapter,
+        None,
+        max_concurrent_reconciles=args.max_concurrent_reconciles,
+        enable_webhook=os.environ.get("ENABLE_WEBHOOKS", "") != "false",
+        syncer_period=args.syncer_period,
+        metrics_port=parse_port(args.metrics_bind_address, 8443),
+    )
+    probe_fn = None
+    try:
+        from ..nodeops.probe import load_library, probe_fn_for_nodeops
+
+        if load_library(required=False) is not None:
+            probe_fn = probe_fn_for_nodeops
+    except Exception:
+        log.warning("gfx950 probe library unavailable; health probe disabled")
+    node_ops = AmdNodeOps(
+        LocalNodeExec(),
+        client=mgr.client,
+        cdi_dir=args.cdi_dir,
+        destructive=args.destructive,
+        probe_fn=probe_fn,
+    )
+    mgr.resource_reconciler.node_ops = node_ops
+    if hasattr(mgr, "syncer"):
+        mgr.syncer.node_ops = node_ops
+
+    mgr.start()
+    log.info("manager started (%d reconcile workers per controller)",
+             args.max_concurrent_reconciles)
+
+    from ..server.api import build_app
+
+    import uvicorn
+
+    app = build_app(mgr.client)
+    server = uvicorn.Server(
+        uvicorn.Config(
+            app,
+            host="0.0.0.0",
+            port=parse_port(args.api_bind_address, 8080),
+            log_level="warning",
+        )
+    )
+
+    stop = threading.Event()
+
+    def handle_signal(signum, frame):
+        log.info("signal %s; shutting down", signum)
+        server.should_exit = True
+        stop.set()
+
+    signal.signal(signal.SIGTERM, handle_signal)
+    signal.signal(signal.SIGINT, handle_signal)
+
+    server.run()  # serves API + healthz/readyz/metrics until signal
+    mgr.stop()
+    if lock_file is not None:
+        lock_file.close()
+    return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

@@ -1,0 +1,151 @@
+"""REST API surface for the standalone operator.
+
+In cluster mode the kube-apiserver owns the CRD objects; in standalone mode
+(no Kubernetes) this server IS the API: apiserver-shaped routes over the
+in-process store, so tooling can drive the operator the same way in both
+modes:
+
+    GET/POST      /apis/cro.hpsys.ibm.ie.com/v1alpha1/{plural}
+    GET/PUT/DELETE /apis/cro.hpsys.ibm.ie.com/v1alpha1/{plural}/{name}
+    PUT           /apis/cro.hpsys.ibm.ie.com/v1alpha1/{plural}/{name}/status
+    GET           /apis/cro.hpsys.ibm.ie.com/v1alpha1/nodes        (+ PUT)
+    GET           /metrics, /healthz, /readyz
+
+Admission (webhook rules + schema validation) runs on every write through
+the store's admission chain, exactly as in-process reconciles see it.
+"""
+
+from __future__ import annotations
+
+from fastapi import FastAPI, HTTPException, Request, Response
+
+from .. import API_VERSION
+from ..api.v1alpha1.types import (
+    ComposabilityRequest,
+    ComposableResource,
+    DeviceTaintRule,
+    Node,
+    ResourceSlice,
+)
+from ..runtime.client import Client
+from ..runtime.errors import (
+    AdmissionDenied,
+    AlreadyExistsError,
+    ConflictError,
+    NotFoundError,
+)
+
+PLURALS = {
+    "composabilityrequests": ComposabilityRequest,
+    "composableresources": ComposableResource,
+    "resourceslices": ResourceSlice,
+    "devicetaintrules": DeviceTaintRule,
+    "nodes": Node,
+}
+
+BASE = "/apis/cro.hpsys.ibm.ie.com/v1alpha1"
+
+
+def _cls(plural: str):
+    cls = PLURALS.get(plural)
+    if cls is None:
+        raise HTTPException(404, f"unknown resource {plural!r}")
+    return cls
+
+
+def _dump(obj) -> dict:
+    return obj.model_dump(by_alias=True)
+
+
+def _http_error(exc: Exception) -> HTTPException:
+    if isinstance(exc, NotFoundError):
+        return HTTPException(404, str(exc))
+    if isinstance(exc, AlreadyExistsError):
+        return HTTPException(409, str(exc))
+    if isinstance(exc, ConflictError):
+        return HTTPException(409, str(exc))
+    if isinstance(exc, AdmissionDenied):
+        return HTTPException(403, str(exc))
+    if isinstance(exc, ValueError):
+        return HTTPException(422, str(exc))
+    return HTTPException(500, str(exc))
+
+
+def build_app(client: Client) -> FastAPI:
+    app = FastAPI(title="cro-amd API", version=API_VERSION)
+
+    @app.get("/healthz")
+    def healthz():
+        return {"status": "ok"}
+
+    @app.get("/readyz")
+    def readyz():
+        return {"status": "ok"}
+
+    @app.get("/metrics")
+    def metrics():
+        import prometheus_client
+
+        return Response(
+            prometheus_client.generate_latest(),
+            media_type=prometheus_client.CONTENT_TYPE_LATEST,
+        )
+
+    @app.get(BASE + "/{plural}")
+    def list_objects(plural: str, labelSelector: str = ""):
+        cls = _cls(plural)
+        labels = None
+        if labelSelector:
+            labels = dict(part.split("=", 1) for part in labelSelector.split(","))
+        items = client.list(cls, labels)
+        return {
+            "apiVersion": API_VERSION,
+            "kind": cls.KIND + "List",
+            "items": [_dump(o) for o in items],
+        }
+
+    @app.get(BASE + "/{plural}/{name:path}")
+    def get_object(plural: str, name: str):
+        cls = _cls(plural)
+        try:
+            return _dump(client.get(cls, name))
+        except Exception as exc:
+            raise _http_error(exc)
+
+    @app.post(BASE + "/{plural}", status_code=201)
+    async def create_object(plural: str, request: Request):
+        cls = _cls(plural)
+        try:
+            obj = cls.model_validate(await request.json())
+            return _dump(client.create(obj))
+        except Exception as exc:
+            raise _http_error(exc)
+
+    @app.put(BASE + "/{plural}/{name:path}")
+    async def update_object(plural: str, name: str, request: Request):
+        cls = _cls(plural)
+        if name.endswith("/status"):
+            name = name[: -len("/status")]
+            try:
+                obj = cls.model_validate(await request.json())
+                obj.metadata.name = name
+                return _dump(client.update_status(obj))
+            except Exception as exc:
+                raise _http_error(exc)
+        try:
+            obj = cls.model_validate(await request.json())
+            obj.metadata.name = name
+            return _dump(client.update(obj))
+        except Exception as exc:
+            raise _http_error(exc)
+
+    @app.delete(BASE + "/{plural}/{name:path}", status_code=202)
+    def delete_object(plural: str, name: str):
+        cls = _cls(plural)
+        try:
+            client.delete(cls, name)
+        except Exception as exc:
+            raise _http_error(exc)
+        return {"status": "deleted"}
+
+    return app
