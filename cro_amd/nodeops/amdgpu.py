@@ -155,14 +155,28 @@ class AmdNodeOps(NodeOps):
     # -- loads -------------------------------------------------------------
 
     def check_no_loads(self, node: str, device_id: Optional[str] = None) -> None:
-        gpu_id = None
+        """Per-device (DRA) or node-wide (DEVICE_PLUGIN) load check.
+
+        "Node-wide" means GPUs *this node can enumerate*: under cgroup
+        device isolation /sys/class/kfd/kfd/proc shows every host process
+        with any KFD context, but processes whose VRAM sits on devices
+        invisible to this node cannot possibly hold ours — counting them
+        (as a naive whole-dir scan would) wedges detach on shared machines.
+        """
         if device_id is not None:
             gpu = self.find_gpu(node, device_id)
             if gpu is None:
                 return  # device already gone — nothing can be loading it
-            gpu_id = gpu.gpu_id
-        pids = gpu_compute_pids(self.execer, node, gpu_id)
-        pids = [p for p in pids if p not in self._own_pids(node, gpu_id)]
+            gpu_ids = [gpu.gpu_id]
+        else:
+            gpu_ids = [g.gpu_id for g in self.enumerate(node)]
+            if not gpu_ids:
+                return
+        pids = gpu_compute_pids(self.execer, node, gpu_ids)
+        own = set()
+        for gid in gpu_ids:
+            own |= self._own_pids(node, gid)
+        pids = [p for p in pids if p not in own]
         if pids:
             scope = f"device {device_id}" if device_id else f"node {node}"
             raise GPULoadsPresent(f"{scope} has active KFD compute processes: {pids}")

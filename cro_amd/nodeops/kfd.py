@@ -305,16 +305,23 @@ def gpu_compute_pids(execer: NodeExec, node: str, gpu_id: Optional[int] = None) 
         if gpu_id is None:
             result.append(int(pid))
             continue
+        if isinstance(gpu_id, (list, tuple, set)):
+            if any(_pid_vram(execer, node, pid, g) > 0 for g in gpu_id):
+                result.append(int(pid))
+            continue
         # per-device attribution: a pid loads THIS gpu iff its vram_<gpu_id>
         # file reads > 0.  A missing/unreadable file means the process has no
         # context on this device (other tenants' processes on a shared node
         # are visible in the proc dir but their per-GPU files are not ours to
         # read) — do NOT count those, or detach wedges forever on shared
         # machines.  The node-wide check (gpu_id=None) stays conservative.
-        try:
-            vram = execer.read_file(node, f"{KFD_PROC}/{pid}/vram_{gpu_id}").strip()
-            if int(vram) > 0:
-                result.append(int(pid))
-        except (FileNotFoundError, PermissionError, OSError, ValueError):
-            continue
+        if _pid_vram(execer, node, pid, gpu_id) > 0:
+            result.append(int(pid))
     return result
+
+
+def _pid_vram(execer: NodeExec, node: str, pid: str, gpu_id: int) -> int:
+    try:
+        return int(execer.read_file(node, f"{KFD_PROC}/{pid}/vram_{gpu_id}").strip())
+    except (FileNotFoundError, PermissionError, OSError, ValueError):
+        return 0

@@ -230,3 +230,17 @@ def test_taint_helpers(client):
     taints.delete_device_taint(client, r)
     taints.delete_device_taint(client, r)  # idempotent
     assert not taints.has_device_taint(client, r)
+
+
+def test_node_wide_loads_only_count_visible_gpus():
+    """Node-wide (DEVICE_PLUGIN) check must ignore other tenants' processes
+    whose VRAM sits on GPUs this node cannot enumerate."""
+    ex = MockNodeExec()
+    kfd_fixture(ex, 1)  # our gpu_id is 1000
+    ex.set_file(NODE, f"{KFD_PROC}/5555/vram_9999", "1048576")  # foreign GPU
+    ex.set_file(NODE, f"{KFD_PROC}/6666/vram_1000", "0")
+    ops = AmdNodeOps(ex)
+    ops.check_no_loads(NODE)  # no load on OUR gpus → passes
+    ex.set_file(NODE, f"{KFD_PROC}/7777/vram_1000", "2048")
+    with pytest.raises(Exception, match="7777"):
+        ops.check_no_loads(NODE)
