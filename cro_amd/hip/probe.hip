@@ -177,13 +177,20 @@ extern "C" int cro_probe_run(int device, struct CroProbeResult* out) {
   hipFree(dA); hipFree(dB); hipFree(dD);
 
   // -- HBM bandwidth -------------------------------------------------------
-  // 1 GiB src + 1 GiB dst; bounded so 8 concurrent probes fit in 288 GB
-  size_t bytes = (size_t)1 << 30;
+  // 256 MiB src + dst, cached per device across probe calls: the probe runs
+  // on EVERY attach, and re-allocating gigabytes each time puts hipMalloc on
+  // the attach-latency path while pinning more of the 288 GB than a health
+  // gate needs.  2 GiB of traffic per measurement is ample signal.
+  static float4* cached_src[64] = {};
+  static float4* cached_dst[64] = {};
+  size_t bytes = (size_t)256 << 20;
   size_t n4 = bytes / sizeof(float4);
-  float4 *src, *dst;
-  CHECK(hipMalloc(&src, bytes));
-  CHECK(hipMalloc(&dst, bytes));
-  CHECK(hipMemset(src, 0x5a, bytes));
+  if (device < 64 && cached_src[device] == nullptr) {
+    CHECK(hipMalloc(&cached_src[device], bytes));
+    CHECK(hipMalloc(&cached_dst[device], bytes));
+    CHECK(hipMemset(cached_src[device], 0x5a, bytes));
+  }
+  float4 *src = cached_src[device], *dst = cached_dst[device];
   hipEvent_t e0, e1;
   CHECK(hipEventCreate(&e0));
   CHECK(hipEventCreate(&e1));
@@ -191,7 +198,7 @@ extern "C" int cro_probe_run(int device, struct CroProbeResult* out) {
   hipLaunchKernelGGL(bw_copy_kernel, grid, block, 0, 0, src, dst, n4);  // warm
   CHECK(hipDeviceSynchronize());
   CHECK(hipEventRecord(e0));
-  const int reps = 4;
+  const int reps = 8;
   for (int i = 0; i < reps; ++i)
     hipLaunchKernelGGL(bw_copy_kernel, grid, block, 0, 0, src, dst, n4);
   CHECK(hipEventRecord(e1));
@@ -199,11 +206,10 @@ extern "C" int cro_probe_run(int device, struct CroProbeResult* out) {
   float ms = 0.f;
   CHECK(hipEventElapsedTime(&ms, e0, e1));
   out->hbm_gbps = (double)(2.0 * bytes * reps) / (ms * 1e6);
-  hipFree(src); hipFree(dst);
 
   // -- bf16 MFMA rate ------------------------------------------------------
   float* sink;
-  const int blocks = 1024, iters = 8192;
+  const int blocks = 1024, iters = 2048;
   CHECK(hipMalloc(&sink, blocks * sizeof(float)));
   hipLaunchKernelGGL(mfma_bf16_rate_kernel, dim3(blocks), dim3(256), 0, 0, sink, 64);
   CHECK(hipDeviceSynchronize());  // warm

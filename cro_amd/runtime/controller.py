@@ -71,8 +71,27 @@ class Controller:
     # -- event intake ------------------------------------------------------
 
     def start(self, store) -> None:
+        self.start_watch(store)
+        self.start_workers()
+
+    def start_watch(self, store) -> None:
+        """Phase 1: subscribe + informer-cache replay.  The manager runs this
+        for EVERY controller before any worker starts (manager.py) — workers
+        of one controller must not advance state machines before sibling
+        controllers are watching, or cross-controller events are lost."""
         kinds = sorted({s.kind for s in self.sources})
         events = store.watch(kinds)
+
+        # replay (controller-runtime parity): every object already in the
+        # store is delivered as a synthetic ADDED event so a restarted
+        # operator resumes mid-state-machine without an external nudge.
+        # Subscribing BEFORE the list means an object created in the gap is
+        # seen twice, never missed — the workqueue dedupes.
+        from .store import WatchEvent
+
+        for kind in kinds:
+            for obj in store.list(kind):
+                self._dispatch(WatchEvent("ADDED", obj))
 
         def pump():
             while not self._stop.is_set():
@@ -86,6 +105,7 @@ class Controller:
         t.start()
         self._threads.append(t)
 
+    def start_workers(self) -> None:
         for i in range(self.workers):
             t = threading.Thread(target=self._worker, name=f"{self.name}-worker-{i}", daemon=True)
             t.start()

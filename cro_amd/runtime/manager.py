@@ -51,8 +51,13 @@ class Manager:
             import prometheus_client
 
             prometheus_client.start_http_server(self._metrics_port)
+        # two-phase start: all watches subscribed (with cache replay) before
+        # any worker runs, so no controller can advance a state machine past
+        # events a sibling controller has not yet subscribed to
         for c in self._controllers:
-            c.start(self.store)
+            c.start_watch(self.store)
+        for c in self._controllers:
+            c.start_workers()
         for period, fn in self._runnables:
             t = threading.Thread(
                 target=self._tick, args=(period, fn), name="runnable", daemon=True

@@ -1,0 +1,119 @@
+"""Operator process tests: entrypoint serving + checkpoint/resume semantics
+(the reference's restart story: all progress lives in object status, a new
+manager resumes mid-state-machine — SURVEY.md §5.4)."""
+
+import os
+import signal
+import socket
+import subprocess
+import sys
+import time
+
+import httpx
+import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def free_port() -> int:
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+@pytest.mark.timeout(120)
+def test_main_serves_and_shuts_down(tmp_path):
+    port = free_port()
+    env = dict(os.environ)
+    env.update({"DEVICE_RESOURCE_TYPE": "DRA", "CDI_PROVIDER_TYPE": "MOCK"})
+    proc = subprocess.Popen(
+        [
+            sys.executable, "-m", "cro_amd.cmd.main",
+            "--api-bind-address", f":{port}",
+            "--metrics-bind-address", f":{free_port()}",
+            "--node", "test-node",
+            "--cdi-dir", str(tmp_path / "cdi"),
+            "--leader-elect", "--leader-elect-lock", str(tmp_path / "leader.lock"),
+        ],
+        cwd=REPO,
+        env=env,
+        stdout=subprocess.PIPE,
+        stderr=subprocess.STDOUT,
+        text=True,
+    )
+    try:
+        base = f"http://127.0.0.1:{port}"
+        deadline = time.monotonic() + 30
+        up = False
+        while time.monotonic() < deadline:
+            try:
+                if httpx.get(base + "/healthz", timeout=1).status_code == 200:
+                    up = True
+                    break
+            except Exception:
+                time.sleep(0.2)
+        assert up, proc.stdout.read() if proc.poll() is not None else "no healthz"
+
+        # API surface is live
+        resp = httpx.get(
+            base + "/apis/cro.hpsys.ibm.ie.com/v1alpha1/composabilityrequests",
+            timeout=5,
+        )
+        assert resp.status_code == 200
+        assert resp.json()["items"] == []
+        # metrics exposed through the API process
+        assert b"cro_reconcile_total" in httpx.get(base + "/metrics", timeout=5).content
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+            proc.wait()
+    assert proc.returncode == 0
+
+
+def test_operator_restart_resumes_state_machine():
+    """A new manager over the same store finishes what the old one started:
+    the CR status IS the checkpoint."""
+    from cro_amd.api.v1alpha1.types import ComposabilityRequest
+    from cro_amd.bench_harness import build_local_stack
+    from cro_amd.controllers import build_manager
+    from tests.conftest import make_request
+
+    stack = build_local_stack(node_name="node0", use_gpu=False)
+    # do NOT start the manager: create the request and hand-advance only the
+    # request controller so children exist but are not yet Online
+    stack.mgr.client.create(make_request("r1", size=2, target_node="node0"))
+    rec = stack.mgr.request_reconciler
+    for _ in range(3):
+        rec.reconcile("r1")
+    mid = stack.mgr.client.get(ComposabilityRequest, "r1")
+    assert mid.status.state == "Updating"  # checkpoint: mid-flight
+
+    # "restart": fresh manager instance over the same store
+    mgr2 = build_manager(stack.mgr.resource_reconciler.adapter, None, store=stack.mgr.store)
+    mgr2.resource_reconciler.node_ops = stack.ops
+    mgr2.start()  # informer-cache replay re-queues every stored object
+    try:
+        assert mgr2.wait_for(
+            lambda: mgr2.client.get(ComposabilityRequest, "r1").status.state == "Running",
+            timeout=10,
+        )
+    finally:
+        mgr2.stop()
+
+
+def test_leader_election_flock(tmp_path):
+    import fcntl
+
+    lock_path = tmp_path / "leader.lock"
+    first = open(lock_path, "w")
+    fcntl.flock(first, fcntl.LOCK_EX)
+    second = open(lock_path, "w")
+    with pytest.raises(BlockingIOError):
+        fcntl.flock(second, fcntl.LOCK_EX | fcntl.LOCK_NB)
+    fcntl.flock(first, fcntl.LOCK_UN)
+    fcntl.flock(second, fcntl.LOCK_EX | fcntl.LOCK_NB)  # now acquirable
