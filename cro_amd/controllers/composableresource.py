@@ -104,6 +104,22 @@ class ComposableResourceReconciler(Reconciler):
             self._set_error(resource, str(exc))
             raise
 
+    def _phase(self, name: str):
+        """Attach-phase span recorded into cro_attach_phase_seconds."""
+        import contextlib
+
+        @contextlib.contextmanager
+        def span():
+            t0 = time.monotonic()
+            try:
+                yield
+            finally:
+                self.metrics.attach_phase_seconds.labels(name).observe(
+                    time.monotonic() - t0
+                )
+
+        return span()
+
     def _set_error(self, resource: ComposableResource, msg: str) -> None:
         """requeueOnErr parity: persist the failure into .status.error
         (composableresource_controller.go:436-446)."""
@@ -170,7 +186,8 @@ class ComposableResourceReconciler(Reconciler):
         node = resource.spec.target_node
         mode = self.adapter.device_resource_type
 
-        self.node_ops.ensure_driver(node)
+        with self._phase("driver_gate"):
+            self.node_ops.ensure_driver(node)
 
         if resource.status.device_id == "":
             t0 = time.monotonic()
@@ -194,7 +211,8 @@ class ComposableResourceReconciler(Reconciler):
                 self.node_ops.check_no_loads(node)
             except GPULoadsPresent as exc:
                 log.warning("gpu loads during attach on %s: %s", node, exc)
-        self.node_ops.refresh_after_attach(node)
+        with self._phase("node_refresh"):
+            self.node_ops.refresh_after_attach(node)
 
         if mode == "DRA":
             visible = self.node_ops.is_visible_dra(node, resource.status.device_id)
@@ -208,8 +226,10 @@ class ComposableResourceReconciler(Reconciler):
         # *Composable Disaggregated Infrastructure* id the fabric handed out
         # (FM detach keys on it, fm/client.go:231-242) — the container-CDI
         # spec name is amd.com/gpu=<device_id>, never stored over it.
-        self.node_ops.write_cdi(node, resource.status.device_id)
-        probe = self.node_ops.health_probe(node, resource.status.device_id)
+        with self._phase("cdi_write"):
+            self.node_ops.write_cdi(node, resource.status.device_id)
+        with self._phase("health_probe"):
+            probe = self.node_ops.health_probe(node, resource.status.device_id)
         if probe is not None and not probe.get("ok", True):
             raise FabricError(f"gfx950 health probe failed: {probe}")
 
@@ -266,7 +286,12 @@ class ComposableResourceReconciler(Reconciler):
             if mode == "DRA":
                 taints.create_device_taint(self.client, resource)
 
-            self.node_ops.drain(node, resource.status.device_id)
+            try:
+                self.node_ops.drain(node, resource.status.device_id)
+            except amdgpu.DrainInProgress:
+                # last-device drains run asynchronously (module unload can
+                # block); poll completion at detach-wait granularity
+                return Result(requeue_after=self.config.detach_invisible_wait)
 
             t0 = time.monotonic()
             try:

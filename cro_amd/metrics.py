@@ -45,6 +45,14 @@ class Metrics:
             "Latency from Detaching entry to device removal completion",
             buckets=_BUCKETS,
         )
+        self.attach_phase_seconds = Histogram(
+            "cro_attach_phase_seconds",
+            "Per-phase attach timing (driver gate, fabric RTT, node refresh, "
+            "CDI write, health probe) — the reference has no per-phase "
+            "tracing (SURVEY.md §5.1)",
+            ["phase"],
+            buckets=_BUCKETS,
+        )
         self.reconcile_total = Counter(
             "cro_reconcile_total",
             "Reconcile invocations by controller and outcome",
@@ -67,6 +75,7 @@ class Metrics:
             for collector in (
                 cls._singleton.attach_to_ready_seconds,
                 cls._singleton.detach_seconds,
+                cls._singleton.attach_phase_seconds,
                 cls._singleton.reconcile_total,
                 cls._singleton.fabric_request_seconds,
                 cls._singleton.devices_online,

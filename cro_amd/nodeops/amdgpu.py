@@ -49,6 +49,10 @@ class GPULoadsPresent(Exception):
     pass
 
 
+class DrainInProgress(Exception):
+    """Last-device drain is running asynchronously; re-check shortly."""
+
+
 class NodeOps:
     """Surface the ComposableResource controller programs against."""
 
@@ -108,6 +112,10 @@ class AmdNodeOps(NodeOps):
         self.probe_fn = probe_fn  # probe_fn(gpu: GPUDevice) -> result dict
         # gpu_id -> our own host pid holding a KFD context there (or None)
         self._self_pid_cache: Dict[int, Optional[int]] = {}
+        # async last-device drains: device_id -> worker thread / error text
+        self._drain_lock = threading.Lock()
+        self._drain_threads: Dict[str, threading.Thread] = {}
+        self._drain_errors: Dict[str, str] = {}
 
     # -- driver ------------------------------------------------------------
 
@@ -237,6 +245,18 @@ class AmdNodeOps(NodeOps):
         no analog), and there is no drain-status query — drain progress is
         tracked in CR status instead.
         """
+        # a previously started async drain for this device?
+        with self._drain_lock:
+            thread = self._drain_threads.get(device_id)
+            if thread is not None:
+                if thread.is_alive():
+                    raise DrainInProgress(f"drain of {device_id} still running")
+                del self._drain_threads[device_id]
+                err = self._drain_errors.pop(device_id, None)
+                if err is not None:
+                    raise ExecError(err)
+                return  # completed successfully
+
         gpu = self.find_gpu(node, device_id)
         if gpu is None:
             return
@@ -245,11 +265,31 @@ class AmdNodeOps(NodeOps):
                 self._sim_detached.add(device_id)
             return
         remaining = [g for g in self.enumerate(node) if g.device_id != device_id]
-        if not remaining:
-            rc, _, err = self.execer.run(node, ["modprobe", "-r", "amdgpu"], timeout=120)
-            if rc != 0:
-                raise ExecError(f"modprobe -r amdgpu failed: {err}", rc=rc, stderr=err)
-        self.execer.write_file(node, f"/sys/bus/pci/devices/{gpu.pci_bdf}/remove", "1")
+        if remaining:
+            self.execer.write_file(node, f"/sys/bus/pci/devices/{gpu.pci_bdf}/remove", "1")
+            return
+
+        # LAST device: module unload + sysfs remove can block for a long
+        # time while KFD tears down, and the remove can stall the writer —
+        # run asynchronously and report progress on re-checks, the pattern
+        # the reference uses for its async sysfs remove (gpus.go:1534-1585)
+        def unload_and_remove():
+            try:
+                rc, _, err = self.execer.run(node, ["modprobe", "-r", "amdgpu"], timeout=120)
+                if rc != 0:
+                    raise ExecError(f"modprobe -r amdgpu failed: {err}", rc=rc, stderr=err)
+                self.execer.write_file(
+                    node, f"/sys/bus/pci/devices/{gpu.pci_bdf}/remove", "1"
+                )
+            except Exception as exc:
+                with self._drain_lock:
+                    self._drain_errors[device_id] = str(exc)
+
+        t = threading.Thread(target=unload_and_remove, name=f"drain-{device_id}", daemon=True)
+        with self._drain_lock:
+            self._drain_threads[device_id] = t
+        t.start()
+        raise DrainInProgress(f"last-device drain of {device_id} started")
 
     def refresh_after_attach(self, node: str) -> None:
         """Make a newly composed device enumerable and published.

@@ -142,19 +142,25 @@ class MockNodeExec(NodeExec):
         with self._lock:
             self.files.pop((node, path), None)
 
-    def set_command(self, argv_or_bin, result: Tuple[int, str, str]) -> None:
+    def set_command(
+        self, argv_or_bin, result: Tuple[int, str, str], delay: float = 0.0
+    ) -> None:
         key = (argv_or_bin,) if isinstance(argv_or_bin, str) else tuple(argv_or_bin)
         with self._lock:
-            self.commands[key] = result
+            self.commands[key] = (result, delay)
 
     def run(self, node: str, argv: List[str], timeout: float = 60.0) -> Tuple[int, str, str]:
         with self._lock:
             self.calls.append(("run", node, tuple(argv)))
-            if tuple(argv) in self.commands:
-                return self.commands[tuple(argv)]
-            if (argv[0],) in self.commands:
-                return self.commands[(argv[0],)]
-        raise ExecError(f"mock: no canned response for {argv}")
+            entry = self.commands.get(tuple(argv)) or self.commands.get((argv[0],))
+        if entry is None:
+            raise ExecError(f"mock: no canned response for {argv}")
+        result, delay = entry
+        if delay:
+            import time
+
+            time.sleep(delay)
+        return result
 
     def read_file(self, node: str, path: str) -> str:
         with self._lock:

@@ -182,11 +182,19 @@ def test_amd_node_ops_destructive_drain_writes_sysfs():
 
 
 def test_amd_node_ops_last_gpu_drain_unloads_module():
+    import time
+
+    from cro_amd.nodeops.amdgpu import DrainInProgress
+
     ex = MockNodeExec()
     ids = kfd_fixture(ex, 1)
     ex.set_command(("modprobe", "-r", "amdgpu"), (0, "", ""))
     ops = AmdNodeOps(ex, destructive=True)
-    ops.drain(NODE, ids[0])
+    # last device → asynchronous module unload + remove
+    with pytest.raises(DrainInProgress):
+        ops.drain(NODE, ids[0])
+    time.sleep(0.05)
+    ops.drain(NODE, ids[0])  # completion check
     assert ("run", NODE, ("modprobe", "-r", "amdgpu")) in ex.calls
     assert ex.files[(NODE, "/sys/bus/pci/devices/0000:03:00.0/remove")] == "1"
 
@@ -244,3 +252,47 @@ def test_node_wide_loads_only_count_visible_gpus():
     ex.set_file(NODE, f"{KFD_PROC}/7777/vram_1000", "2048")
     with pytest.raises(Exception, match="7777"):
         ops.check_no_loads(NODE)
+
+
+def test_last_gpu_drain_is_async():
+    """Last-device drain runs asynchronously: first call raises
+    DrainInProgress, completion is reported on re-check (the reference's
+    async sysfs-remove pattern, gpus.go:1534-1585)."""
+    import time
+
+    from cro_amd.nodeops.amdgpu import DrainInProgress
+
+    ex = MockNodeExec()
+    ids = kfd_fixture(ex, 1)
+    ex.set_command(("modprobe", "-r", "amdgpu"), (0, "", ""), delay=0.1)
+    ops = AmdNodeOps(ex, destructive=True)
+    with pytest.raises(DrainInProgress):
+        ops.drain(NODE, ids[0])
+    with pytest.raises(DrainInProgress):
+        ops.drain(NODE, ids[0])  # still unloading
+    time.sleep(0.15)
+    ops.drain(NODE, ids[0])  # completed
+    assert ex.files[(NODE, "/sys/bus/pci/devices/0000:03:00.0/remove")] == "1"
+
+
+def test_last_gpu_drain_async_error_surfaces():
+    import time
+
+    from cro_amd.nodeops.amdgpu import DrainInProgress
+    from cro_amd.nodeops.execs import ExecError
+
+    ex = MockNodeExec()
+    ids = kfd_fixture(ex, 1)
+    ex.set_command(("modprobe", "-r", "amdgpu"), (1, "", "module in use"), delay=0.05)
+    ops = AmdNodeOps(ex, destructive=True)
+    with pytest.raises(DrainInProgress):
+        ops.drain(NODE, ids[0])
+    time.sleep(0.1)
+    with pytest.raises(ExecError, match="module in use"):
+        ops.drain(NODE, ids[0])
+    # after the error is consumed, drain can be retried from scratch
+    ex.set_command(("modprobe", "-r", "amdgpu"), (0, "", ""))
+    with pytest.raises(DrainInProgress):
+        ops.drain(NODE, ids[0])
+    time.sleep(0.05)
+    ops.drain(NODE, ids[0])
