@@ -1,0 +1,175 @@
+"""croctl — kubectl-style CLI for the standalone cro-amd API.
+
+    croctl get composabilityrequests [NAME]
+    croctl get composableresources
+    croctl describe composabilityrequests NAME
+    croctl apply -f request.yaml
+    croctl delete composabilityrequests NAME
+    croctl scale composabilityrequests NAME --size N
+
+Server selection: --server or CRO_SERVER (default http://127.0.0.1:8080).
+"""
+
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+
+import httpx
+import yaml
+
+BASE = "/apis/cro.hpsys.ibm.ie.com/v1alpha1"
+
+_COLUMNS = {
+    "composabilityrequests": (
+        ("NAME", lambda o: o["metadata"]["name"]),
+        ("TYPE", lambda o: o["spec"]["resource"]["type"]),
+        ("MODEL", lambda o: o["spec"]["resource"]["model"]),
+        ("SIZE", lambda o: str(o["spec"]["resource"]["size"])),
+        ("STATE", lambda o: o["status"]["state"] or "<none>"),
+        ("DEVICES", lambda o: str(len(o["status"].get("resources", {})))),
+    ),
+    "composableresources": (
+        ("NAME", lambda o: o["metadata"]["name"]),
+        ("TYPE", lambda o: o["spec"]["type"]),
+        ("NODE", lambda o: o["spec"]["target_node"]),
+        ("STATE", lambda o: o["status"]["state"] or "<none>"),
+        ("DEVICE", lambda o: o["status"].get("device_id", "")),
+        ("ERROR", lambda o: (o["status"].get("error", "") or "")[:40]),
+    ),
+}
+_DEFAULT_COLUMNS = (
+    ("NAME", lambda o: o["metadata"]["name"]),
+    ("KIND", lambda o: o.get("kind", "")),
+)
+
+
+def _print_table(plural: str, items: list) -> None:
+    columns = _COLUMNS.get(plural, _DEFAULT_COLUMNS)
+    rows = [[fn(o) for _, fn in columns] for o in items]
+    headers = [h for h, _ in columns]
+    widths = [
+        max(len(headers[i]), *(len(r[i]) for r in rows)) if rows else len(headers[i])
+        for i in range(len(headers))
+    ]
+    print("  ".join(h.ljust(w) for h, w in zip(headers, widths)))
+    for r in rows:
+        print("  ".join(c.ljust(w) for c, w in zip(r, widths)))
+
+
+def main(argv=None, client: httpx.Client = None) -> int:
+    p = argparse.ArgumentParser(prog="croctl")
+    p.add_argument("--server", default=os.environ.get("CRO_SERVER", "http://127.0.0.1:8080"))
+    p.add_argument("-o", "--output", choices=["table", "yaml", "json"], default="table")
+    sub = p.add_subparsers(dest="command", required=True)
+
+    g = sub.add_parser("get")
+    g.add_argument("plural")
+    g.add_argument("name", nargs="?")
+
+    d = sub.add_parser("describe")
+    d.add_argument("plural")
+    d.add_argument("name")
+
+    a = sub.add_parser("apply")
+    a.add_argument("-f", "--filename", required=True)
+
+    rm = sub.add_parser("delete")
+    rm.add_argument("plural")
+    rm.add_argument("name")
+
+    sc = sub.add_parser("scale")
+    sc.add_argument("plural")
+    sc.add_argument("name")
+    sc.add_argument("--size", type=int, required=True)
+
+    args = p.parse_args(argv)
+    http = client or httpx.Client(base_url=args.server, timeout=30)
+
+    def fail(resp) -> int:
+        print(f"error: {resp.status_code}: {resp.text}", file=sys.stderr)
+        return 1
+
+    if args.command == "get":
+        if args.name:
+            resp = http.get(f"{BASE}/{args.plural}/{args.name}")
+            if resp.status_code != 200:
+                return fail(resp)
+            items = [resp.json()]
+        else:
+            resp = http.get(f"{BASE}/{args.plural}")
+            if resp.status_code != 200:
+                return fail(resp)
+            items = resp.json()["items"]
+        if args.output == "json":
+            print(json.dumps(items, indent=2))
+        elif args.output == "yaml":
+            print(yaml.safe_dump(items, sort_keys=False))
+        else:
+            _print_table(args.plural, items)
+        return 0
+
+    if args.command == "describe":
+        resp = http.get(f"{BASE}/{args.plural}/{args.name}")
+        if resp.status_code != 200:
+            return fail(resp)
+        print(yaml.safe_dump(resp.json(), sort_keys=False))
+        return 0
+
+    def update_with_conflict_retry(plural: str, name: str, mutate) -> httpx.Response:
+        """GET-mutate-PUT with optimistic-concurrency retries: concurrent
+        reconciles bump resourceVersion between our read and write."""
+        resp = None
+        for _ in range(8):
+            current = http.get(f"{BASE}/{plural}/{name}")
+            if current.status_code != 200:
+                return current
+            obj = current.json()
+            mutate(obj)
+            resp = http.put(f"{BASE}/{plural}/{name}", json=obj)
+            if resp.status_code != 409:
+                return resp
+        return resp
+
+    if args.command == "apply":
+        with open(args.filename) as f:
+            obj = yaml.safe_load(f)
+        plural = obj["kind"].lower() + "s"
+        name = obj["metadata"]["name"]
+        if http.get(f"{BASE}/{plural}/{name}").status_code == 200:
+            resp = update_with_conflict_retry(
+                plural, name, lambda cur: cur.update({"spec": obj["spec"]})
+            )
+            verb = "configured"
+        else:
+            resp = http.post(f"{BASE}/{plural}", json=obj)
+            verb = "created"
+        if resp.status_code not in (200, 201):
+            return fail(resp)
+        print(f"{plural}/{name} {verb}")
+        return 0
+
+    if args.command == "delete":
+        resp = http.delete(f"{BASE}/{args.plural}/{args.name}")
+        if resp.status_code != 202:
+            return fail(resp)
+        print(f"{args.plural}/{args.name} deleted")
+        return 0
+
+    if args.command == "scale":
+        def set_size(obj):
+            obj["spec"]["resource"]["size"] = args.size
+
+        resp = update_with_conflict_retry(args.plural, args.name, set_size)
+        if resp is None or resp.status_code != 200:
+            return fail(resp)
+        print(f"{args.plural}/{args.name} scaled to {args.size}")
+        return 0
+
+    return 2
+
+
+if __name__ == "__main__":
+    sys.exit(main())

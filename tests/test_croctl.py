@@ -1,0 +1,102 @@
+"""croctl CLI tests against an in-process API server."""
+
+import pytest
+import yaml
+from fastapi.testclient import TestClient
+
+from cro_amd.bench_harness import build_local_stack
+from cro_amd.cmd.croctl import main
+from cro_amd.server.api import build_app
+
+
+@pytest.fixture
+def cli(capsys):
+    stack = build_local_stack(node_name="node0", use_gpu=False)
+    stack.mgr.start()
+    http = TestClient(build_app(stack.mgr.client))
+
+    def run(*args):
+        rc = main(list(args), client=http)
+        out = capsys.readouterr()
+        return rc, out.out, out.err
+
+    yield run, stack
+    stack.mgr.stop()
+
+
+def write_request(tmp_path, name="r1", size=1):
+    path = tmp_path / f"{name}.yaml"
+    path.write_text(
+        yaml.safe_dump(
+            {
+                "apiVersion": "cro.hpsys.ibm.ie.com/v1alpha1",
+                "kind": "ComposabilityRequest",
+                "metadata": {"name": name},
+                "spec": {
+                    "resource": {
+                        "type": "gpu", "model": "mi355x", "size": size,
+                        "target_node": "node0",
+                    }
+                },
+            }
+        )
+    )
+    return str(path)
+
+
+def test_apply_get_scale_delete(cli, tmp_path):
+    run, stack = cli
+    rc, out, _ = run("apply", "-f", write_request(tmp_path))
+    assert rc == 0 and "created" in out
+
+    assert stack.mgr.wait_for(
+        lambda: run("get", "composabilityrequests", "r1")[1].count("Running") > 0,
+        timeout=10,
+    )
+    rc, out, _ = run("get", "composabilityrequests")
+    assert rc == 0
+    assert "NAME" in out and "r1" in out and "mi355x" in out
+
+    rc, out, _ = run("scale", "composabilityrequests", "r1", "--size", "3")
+    assert rc == 0
+    assert stack.mgr.wait_for(
+        lambda: "DEVICES" in run("get", "composabilityrequests", "r1")[1]
+        and "3" in run("get", "composabilityrequests", "r1")[1].split("\n")[1],
+        timeout=10,
+    )
+
+    rc, out, _ = run("describe", "composabilityrequests", "r1")
+    assert rc == 0
+    described = yaml.safe_load(out)
+    assert described["spec"]["resource"]["size"] == 3
+
+    rc, out, _ = run("delete", "composabilityrequests", "r1")
+    assert rc == 0
+    assert stack.mgr.wait_for(
+        lambda: run("get", "composabilityrequests", "r1")[0] == 1, timeout=10
+    )
+
+
+def test_apply_update_existing(cli, tmp_path):
+    run, stack = cli
+    run("apply", "-f", write_request(tmp_path, size=1))
+    rc, out, _ = run("apply", "-f", write_request(tmp_path, size=2))
+    assert rc == 0 and "configured" in out
+
+
+def test_get_composableresources_table(cli, tmp_path):
+    run, stack = cli
+    run("apply", "-f", write_request(tmp_path))
+    assert stack.mgr.wait_for(
+        lambda: "Online" in run("get", "composableresources")[1], timeout=10
+    )
+    rc, out, _ = run("get", "composableresources")
+    assert "node0" in out and "GPU-" in out
+
+
+def test_error_paths(cli):
+    run, _ = cli
+    rc, _, err = run("get", "composabilityrequests", "nope")
+    assert rc == 1 and "404" in err
+    rc, _, err = run("delete", "composabilityrequests", "nope")
+    assert rc == 1
