@@ -112,6 +112,24 @@ def main() -> int:
     barrier()
     t_end = time.monotonic()
     rec1 = reconcile_count(stack)
+
+    # optional attach-phase breakdown (CRO_BENCH_PHASES=<path>) — goes to a
+    # side file so rank 0's stdout stays the single contract JSON line
+    phases_path = os.environ.get("CRO_BENCH_PHASES", "")
+    if phases_path and rank == 0:
+        phases = {}
+        for metric in stack.mgr.metrics.attach_phase_seconds.collect():
+            for s in metric.samples:
+                if s.name.endswith("_sum"):
+                    phases.setdefault(s.labels["phase"], {})["sum_s"] = s.value
+                elif s.name.endswith("_count"):
+                    phases.setdefault(s.labels["phase"], {})["count"] = s.value
+        for v in phases.values():
+            if v.get("count"):
+                v["avg_ms"] = round(v["sum_s"] * 1e3 / v["count"], 3)
+        with open(phases_path, "w") as f:
+            json.dump(phases, f, indent=2)
+
     stack.mgr.stop()
 
     elapsed = t_end - t_start
