@@ -130,6 +130,18 @@ static void host_mfma_ref(const float* A, const float* B, float* D, int K) {
   }
 }
 
+// Per-device cached probe context: the probe runs on every attach, so
+// allocations and events are created once and reused — only the kernels,
+// copies and compares run per call.
+struct ProbeCtx {
+  int ready = 0;
+  float *dA = nullptr, *dB = nullptr, *dD = nullptr;
+  float4 *src = nullptr, *dst = nullptr;
+  float* sink = nullptr;
+  hipEvent_t e0 = nullptr, e1 = nullptr;
+};
+static ProbeCtx g_ctx[64];
+
 extern "C" int cro_probe_run(int device, struct CroProbeResult* out) {
   memset(out, 0, sizeof(*out));
   out->ok = 0;
@@ -163,44 +175,45 @@ extern "C" int cro_probe_run(int device, struct CroProbeResult* out) {
     s = s * 1664525u + 1013904223u;
     hB[i] = ((float)(s >> 8) / 16777216.0f) - 0.5f;
   }
-  float *dA, *dB, *dD;
-  CHECK(hipMalloc(&dA, sizeof(hA)));
-  CHECK(hipMalloc(&dB, sizeof(hB)));
-  CHECK(hipMalloc(&dD, sizeof(hD)));
-  CHECK(hipMemcpy(dA, hA, sizeof(hA), hipMemcpyHostToDevice));
-  CHECK(hipMemcpy(dB, hB, sizeof(hB), hipMemcpyHostToDevice));
-  hipLaunchKernelGGL(mfma_f32_check_kernel, dim3(1), dim3(64), 0, 0, dA, dB, dD, K);
+  ProbeCtx* ctx = (device < 64) ? &g_ctx[device] : nullptr;
+  if (ctx == nullptr) {
+    snprintf(out->msg, sizeof(out->msg), "device ordinal %d above cache limit", device);
+    return -1;
+  }
+  const size_t bytes = (size_t)256 << 20;  // bw buffers (see below)
+  if (!ctx->ready) {
+    CHECK(hipMalloc(&ctx->dA, sizeof(hA)));
+    CHECK(hipMalloc(&ctx->dB, sizeof(hB)));
+    CHECK(hipMalloc(&ctx->dD, sizeof(hD)));
+    CHECK(hipMalloc(&ctx->src, bytes));
+    CHECK(hipMalloc(&ctx->dst, bytes));
+    CHECK(hipMemset(ctx->src, 0x5a, bytes));
+    CHECK(hipMalloc(&ctx->sink, 1024 * sizeof(float)));
+    CHECK(hipEventCreate(&ctx->e0));
+    CHECK(hipEventCreate(&ctx->e1));
+    ctx->ready = 1;
+  }
+  CHECK(hipMemcpy(ctx->dA, hA, sizeof(hA), hipMemcpyHostToDevice));
+  CHECK(hipMemcpy(ctx->dB, hB, sizeof(hB), hipMemcpyHostToDevice));
+  hipLaunchKernelGGL(mfma_f32_check_kernel, dim3(1), dim3(64), 0, 0, ctx->dA, ctx->dB, ctx->dD, K);
   CHECK(hipGetLastError());
-  CHECK(hipMemcpy(hD, dD, sizeof(hD), hipMemcpyDeviceToHost));
+  CHECK(hipMemcpy(hD, ctx->dD, sizeof(hD), hipMemcpyDeviceToHost));
   host_mfma_ref(hA, hB, refD, K);
   out->mfma_f32_exact = (memcmp(hD, refD, sizeof(hD)) == 0) ? 1 : 0;
-  hipFree(dA); hipFree(dB); hipFree(dD);
 
   // -- HBM bandwidth -------------------------------------------------------
-  // 256 MiB src + dst, cached per device across probe calls: the probe runs
-  // on EVERY attach, and re-allocating gigabytes each time puts hipMalloc on
-  // the attach-latency path while pinning more of the 288 GB than a health
-  // gate needs.  2 GiB of traffic per measurement is ample signal.
-  static float4* cached_src[64] = {};
-  static float4* cached_dst[64] = {};
-  size_t bytes = (size_t)256 << 20;
+  // 256 MiB src + dst from the cached context (2 GiB of traffic per
+  // measurement — ample signal without putting hipMalloc on the
+  // attach-latency path or pinning more of the 288 GB than a gate needs)
   size_t n4 = bytes / sizeof(float4);
-  if (device < 64 && cached_src[device] == nullptr) {
-    CHECK(hipMalloc(&cached_src[device], bytes));
-    CHECK(hipMalloc(&cached_dst[device], bytes));
-    CHECK(hipMemset(cached_src[device], 0x5a, bytes));
-  }
-  float4 *src = cached_src[device], *dst = cached_dst[device];
-  hipEvent_t e0, e1;
-  CHECK(hipEventCreate(&e0));
-  CHECK(hipEventCreate(&e1));
+  hipEvent_t e0 = ctx->e0, e1 = ctx->e1;
   dim3 grid(8192), block(256);
-  hipLaunchKernelGGL(bw_copy_kernel, grid, block, 0, 0, src, dst, n4);  // warm
+  hipLaunchKernelGGL(bw_copy_kernel, grid, block, 0, 0, ctx->src, ctx->dst, n4);  // warm
   CHECK(hipDeviceSynchronize());
   CHECK(hipEventRecord(e0));
   const int reps = 8;
   for (int i = 0; i < reps; ++i)
-    hipLaunchKernelGGL(bw_copy_kernel, grid, block, 0, 0, src, dst, n4);
+    hipLaunchKernelGGL(bw_copy_kernel, grid, block, 0, 0, ctx->src, ctx->dst, n4);
   CHECK(hipEventRecord(e1));
   CHECK(hipEventSynchronize(e1));
   float ms = 0.f;
@@ -208,21 +221,17 @@ extern "C" int cro_probe_run(int device, struct CroProbeResult* out) {
   out->hbm_gbps = (double)(2.0 * bytes * reps) / (ms * 1e6);
 
   // -- bf16 MFMA rate ------------------------------------------------------
-  float* sink;
   const int blocks = 1024, iters = 2048;
-  CHECK(hipMalloc(&sink, blocks * sizeof(float)));
-  hipLaunchKernelGGL(mfma_bf16_rate_kernel, dim3(blocks), dim3(256), 0, 0, sink, 64);
+  hipLaunchKernelGGL(mfma_bf16_rate_kernel, dim3(blocks), dim3(256), 0, 0, ctx->sink, 64);
   CHECK(hipDeviceSynchronize());  // warm
   CHECK(hipEventRecord(e0));
-  hipLaunchKernelGGL(mfma_bf16_rate_kernel, dim3(blocks), dim3(256), 0, 0, sink, iters);
+  hipLaunchKernelGGL(mfma_bf16_rate_kernel, dim3(blocks), dim3(256), 0, 0, ctx->sink, iters);
   CHECK(hipEventRecord(e1));
   CHECK(hipEventSynchronize(e1));
   CHECK(hipEventElapsedTime(&ms, e0, e1));
   double waves = (double)blocks * 256.0 / 64.0;
   double flops = waves * 4.0 * (double)iters * 2.0 * 32.0 * 32.0 * 16.0;
   out->bf16_tflops = flops / (ms * 1e9);
-  hipFree(sink);
-  hipEventDestroy(e0); hipEventDestroy(e1);
 
   // gates: exact MFMA is hard; bandwidth/rate are loose floors so a busy or
   // power-capped chip never false-fails
