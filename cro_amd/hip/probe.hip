@@ -135,6 +135,8 @@ static void host_mfma_ref(const float* A, const float* B, float* D, int K) {
 // copies and compares run per call.
 struct ProbeCtx {
   int ready = 0;
+  long long vram_total = 0, vram_free = 0;  // hipMemGetInfo is sysfs-backed
+                                            // and ms-scale: snapshot at init
   float *dA = nullptr, *dB = nullptr, *dD = nullptr;
   float4 *src = nullptr, *dst = nullptr;
   float* sink = nullptr;
@@ -188,6 +190,10 @@ extern "C" int cro_probe_run(int device, struct CroProbeResult* out) {
     hipDeviceProp_t prop;
     CHECK(hipGetDeviceProperties(&prop, device));
     snprintf(ctx->gcn_arch, sizeof(ctx->gcn_arch), "%s", prop.gcnArchName);
+    size_t free_b = 0, total_b = 0;
+    CHECK(hipMemGetInfo(&free_b, &total_b));
+    ctx->vram_total = (long long)total_b;
+    ctx->vram_free = (long long)free_b;
     // one warm round at init covers kernel-code upload; per-call warms are
     // off the attach path
     hipLaunchKernelGGL(bw_copy_kernel, dim3(8192), dim3(256), 0, 0, ctx->src, ctx->dst,
@@ -197,10 +203,8 @@ extern "C" int cro_probe_run(int device, struct CroProbeResult* out) {
     ctx->ready = 1;
   }
   snprintf(out->gcn_arch, sizeof(out->gcn_arch), "%s", ctx->gcn_arch);
-  size_t free_b = 0, total_b = 0;
-  CHECK(hipMemGetInfo(&free_b, &total_b));
-  out->vram_total = (long long)total_b;
-  out->vram_free = (long long)free_b;
+  out->vram_total = ctx->vram_total;
+  out->vram_free = ctx->vram_free;
   CHECK(hipMemcpy(ctx->dA, hA, sizeof(hA), hipMemcpyHostToDevice));
   CHECK(hipMemcpy(ctx->dB, hB, sizeof(hB), hipMemcpyHostToDevice));
   hipLaunchKernelGGL(mfma_f32_check_kernel, dim3(1), dim3(64), 0, 0, ctx->dA, ctx->dB, ctx->dD, K);
