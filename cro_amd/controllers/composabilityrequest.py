@@ -20,7 +20,6 @@ watch predicate the transition is usually event-driven anyway.
 from __future__ import annotations
 
 import logging
-import time
 from dataclasses import dataclass
 from typing import Dict, List, Optional
 
