@@ -60,6 +60,8 @@ def build_local_stack(
     fabric_config: Optional[MockFabricConfig] = None,
     enable_probe: bool = True,
     max_concurrent_reconciles: int = 8,
+    syncer_period: Optional[float] = 1.0,
+    syncer_grace: float = 10.0,
 ) -> LocalStack:
     if use_gpu is None:
         use_gpu = gpu_available()
@@ -95,6 +97,8 @@ def build_local_stack(
             resource_config=ReconcileConfig(),
             request_config=RequestReconcileConfig(),
             max_concurrent_reconciles=max_concurrent_reconciles,
+            syncer_period=syncer_period,
+            syncer_grace=syncer_grace,
         )
         ops = AmdNodeOps(
             execer,
@@ -105,6 +109,8 @@ def build_local_stack(
             probe_fn=probe_fn,
         )
         mgr.resource_reconciler.node_ops = ops
+        if hasattr(mgr, "syncer"):
+            mgr.syncer.node_ops = ops
 
         # bridge fabric composition → simulated hot-add (a real fabric would
         # make the device appear on the PCIe bus; here the silicon is already
@@ -126,9 +132,13 @@ def build_local_stack(
             resource_config=ReconcileConfig(),
             request_config=RequestReconcileConfig(),
             max_concurrent_reconciles=max_concurrent_reconciles,
+            syncer_period=syncer_period,
+            syncer_grace=syncer_grace,
         )
         ops = MockNodeOps(client=mgr.client)
         mgr.resource_reconciler.node_ops = ops
+        if hasattr(mgr, "syncer"):
+            mgr.syncer.node_ops = ops
         ops.set_driver(node_name, True)
         orig_add = fabric.add_resource
 

@@ -104,6 +104,21 @@ class ComposableResourceReconciler(Reconciler):
             self._set_error(resource, str(exc))
             raise
 
+    def _persist_device_identity(
+        self, name: str, device_id: str, cdi_device_id: str
+    ) -> ComposableResource:
+        last_exc = None
+        for _ in range(8):
+            try:
+                fresh = self.client.get(ComposableResource, name)
+                fresh.status.error = ""
+                fresh.status.device_id = device_id
+                fresh.status.cdi_device_id = cdi_device_id
+                return self.client.update_status(fresh)
+            except ConflictError as exc:
+                last_exc = exc
+        raise last_exc
+
     def _phase(self, name: str):
         """Attach-phase span recorded into cro_attach_phase_seconds."""
         import contextlib
@@ -199,10 +214,15 @@ class ComposableResourceReconciler(Reconciler):
                 self.metrics.fabric_request_seconds.labels(
                     self.adapter.provider.name, "add"
                 ).observe(time.monotonic() - t0)
-            resource.status.error = ""
-            resource.status.device_id = device_id
-            resource.status.cdi_device_id = cdi_device_id
-            resource = self.client.update_status(resource)
+            # Persist the fabric-assigned identity with conflict-retry: a
+            # concurrent deletion bumps the RV, and dropping this write
+            # would leak the attached device (the CR would reach Deleting
+            # with device_id "" while the fabric holds the attachment —
+            # the upstream syncer would only repair it after the grace
+            # period).  The identity is correct regardless of what raced.
+            resource = self._persist_device_identity(
+                resource.metadata.name, device_id, cdi_device_id
+            )
 
         if mode == "DEVICE_PLUGIN":
             # load check is advisory on attach (reference logs and continues,
