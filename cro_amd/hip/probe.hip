@@ -23,8 +23,15 @@
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
+#include <chrono>
 #include <cstdio>
 #include <cstring>
+
+static double now_ms() {
+  return std::chrono::duration<double, std::milli>(
+             std::chrono::steady_clock::now().time_since_epoch())
+      .count();
+}
 
 #define CHECK(expr)                                                          \
   do {                                                                       \
@@ -50,6 +57,10 @@ struct CroProbeResult {
   double bf16_tflops;       // dense bf16 MFMA issue rate
   long long vram_total;     // bytes
   long long vram_free;      // bytes
+  double t_setup_ms;        // host wall per section (diagnosing the
+  double t_mfma_ms;         // attach-latency contribution of the probe)
+  double t_bw_ms;
+  double t_bf16_ms;
   char gcn_arch[64];
   char msg[256];
 };
@@ -149,6 +160,7 @@ static ProbeCtx g_ctx[64];
 extern "C" int cro_probe_run(int device, struct CroProbeResult* out) {
   memset(out, 0, sizeof(*out));
   out->ok = 0;
+  double t0 = now_ms();
 
   int count = 0;
   CHECK(hipGetDeviceCount(&count));
@@ -160,6 +172,8 @@ extern "C" int cro_probe_run(int device, struct CroProbeResult* out) {
   CHECK(hipSetDevice(device));
 
   // -- exact f32 MFMA ------------------------------------------------------
+  out->t_setup_ms = now_ms() - t0;
+  t0 = now_ms();
   const int K = 64;
   float hA[16 * K], hB[K * 16], hD[256], refD[256];
   unsigned s = 0x9e3779b9u;
@@ -213,6 +227,8 @@ extern "C" int cro_probe_run(int device, struct CroProbeResult* out) {
   host_mfma_ref(hA, hB, refD, K);
   out->mfma_f32_exact = (memcmp(hD, refD, sizeof(hD)) == 0) ? 1 : 0;
 
+  out->t_mfma_ms = now_ms() - t0;
+  t0 = now_ms();
   // -- HBM bandwidth -------------------------------------------------------
   // 256 MiB src + dst from the cached context (2 GiB of traffic per
   // measurement — ample signal without putting hipMalloc on the
@@ -230,6 +246,8 @@ extern "C" int cro_probe_run(int device, struct CroProbeResult* out) {
   CHECK(hipEventElapsedTime(&ms, e0, e1));
   out->hbm_gbps = (double)(2.0 * bytes * reps) / (ms * 1e6);
 
+  out->t_bw_ms = now_ms() - t0;
+  t0 = now_ms();
   // -- bf16 MFMA rate ------------------------------------------------------
   const int blocks = 1024, iters = 2048;
   CHECK(hipEventRecord(e0));
@@ -240,6 +258,7 @@ extern "C" int cro_probe_run(int device, struct CroProbeResult* out) {
   double waves = (double)blocks * 256.0 / 64.0;
   double flops = waves * 4.0 * (double)iters * 2.0 * 32.0 * 32.0 * 16.0;
   out->bf16_tflops = flops / (ms * 1e9);
+  out->t_bf16_ms = now_ms() - t0;
 
   // gates: exact MFMA is hard; bandwidth/rate are loose floors so a busy or
   // power-capped chip never false-fails

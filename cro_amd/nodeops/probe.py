@@ -22,6 +22,10 @@ class CroProbeResult(ctypes.Structure):
         ("bf16_tflops", ctypes.c_double),
         ("vram_total", ctypes.c_longlong),
         ("vram_free", ctypes.c_longlong),
+        ("t_setup_ms", ctypes.c_double),
+        ("t_mfma_ms", ctypes.c_double),
+        ("t_bw_ms", ctypes.c_double),
+        ("t_bf16_ms", ctypes.c_double),
         ("gcn_arch", ctypes.c_char * 64),
         ("msg", ctypes.c_char * 256),
     ]
@@ -82,16 +86,23 @@ def device_count() -> int:
     return max(lib.cro_probe_device_count(), 0)
 
 
+_bdf_cache = {}
+
+
 def hip_device_for_bdf(pci_bdf: str) -> Optional[int]:
-    """Map a PCI DBDF (from KFD topology) to a HIP device ordinal."""
+    """Map a PCI DBDF (from KFD topology) to a HIP device ordinal (cached —
+    hipDeviceGetPCIBusId is sysfs-backed and the probe runs per attach)."""
+    want = pci_bdf.lower()
+    if want in _bdf_cache:
+        return _bdf_cache[want]
     lib = load_library()
     if lib is None:
         return None
     buf = ctypes.create_string_buffer(64)
-    want = pci_bdf.lower()
     for dev in range(device_count()):
         if lib.cro_probe_pci_bus_id(dev, buf, 64) == 0:
             if buf.value.decode().lower().startswith(want.rsplit(".", 1)[0]):
+                _bdf_cache[want] = dev
                 return dev
     return None
 
@@ -108,6 +119,10 @@ def run_probe(device: int = 0) -> dict:
         "bf16_tflops": res.bf16_tflops,
         "vram_total": res.vram_total,
         "vram_free": res.vram_free,
+        "t_setup_ms": res.t_setup_ms,
+        "t_mfma_ms": res.t_mfma_ms,
+        "t_bw_ms": res.t_bw_ms,
+        "t_bf16_ms": res.t_bf16_ms,
         "gcn_arch": res.gcn_arch.decode(errors="replace"),
         "msg": res.msg.decode(errors="replace"),
     }
