@@ -116,6 +116,12 @@ class AmdNodeOps(NodeOps):
         self._drain_lock = threading.Lock()
         self._drain_threads: Dict[str, threading.Thread] = {}
         self._drain_errors: Dict[str, str] = {}
+        # short-TTL enumeration cache: one attach touches the topology 4+
+        # times (refresh/visibility/CDI/probe) and each full KFD sysfs scan
+        # costs ~2 ms; lifecycle mutations invalidate explicitly
+        self.enum_cache_ttl = 0.025
+        self._enum_cache: Dict[str, tuple] = {}  # node -> (monotonic, gpus)
+        self._enum_lock = threading.Lock()
 
     # -- driver ------------------------------------------------------------
 
@@ -131,14 +137,28 @@ class AmdNodeOps(NodeOps):
     # -- enumeration / visibility -----------------------------------------
 
     def enumerate(self, node: str) -> List[GPUDevice]:
-        try:
-            gpus = enumerate_gpus(self.execer, node)
-        except ExecError:
-            gpus = enumerate_gpus_amdsmi(self.execer, node)
+        now = time.monotonic()
+        with self._enum_lock:
+            cached = self._enum_cache.get(node)
+            if cached is not None and now - cached[0] < self.enum_cache_ttl:
+                gpus = cached[1]
+            else:
+                gpus = None
+        if gpus is None:
+            try:
+                gpus = enumerate_gpus(self.execer, node)
+            except ExecError:
+                gpus = enumerate_gpus_amdsmi(self.execer, node)
+            with self._enum_lock:
+                self._enum_cache[node] = (now, gpus)
         if self.destructive:
-            return gpus
+            return list(gpus)
         with self._sim_lock:
             return [g for g in gpus if g.device_id not in self._sim_detached]
+
+    def _invalidate_enum(self, node: str) -> None:
+        with self._enum_lock:
+            self._enum_cache.pop(node, None)
 
     def find_gpu(self, node: str, device_id: str) -> Optional[GPUDevice]:
         for g in self.enumerate(node):
@@ -263,10 +283,12 @@ class AmdNodeOps(NodeOps):
         if not self.destructive:
             with self._sim_lock:
                 self._sim_detached.add(device_id)
+            self._invalidate_enum(node)
             return
         remaining = [g for g in self.enumerate(node) if g.device_id != device_id]
         if remaining:
             self.execer.write_file(node, f"/sys/bus/pci/devices/{gpu.pci_bdf}/remove", "1")
+            self._invalidate_enum(node)
             return
 
         # LAST device: module unload + sysfs remove can block for a long
@@ -281,6 +303,7 @@ class AmdNodeOps(NodeOps):
                 self.execer.write_file(
                     node, f"/sys/bus/pci/devices/{gpu.pci_bdf}/remove", "1"
                 )
+                self._invalidate_enum(node)
             except Exception as exc:
                 with self._drain_lock:
                     self._drain_errors[device_id] = str(exc)
@@ -304,6 +327,7 @@ class AmdNodeOps(NodeOps):
             self.execer.write_file(node, "/sys/bus/pci/rescan", "1")
         # non-destructive mode: simulate_compose() already restored the
         # device into the enumerable set; rescan is a no-op by design
+        self._invalidate_enum(node)  # topology just changed
         for name, hook in self.restart_hooks.items():
             try:
                 hook(node)
@@ -316,6 +340,7 @@ class AmdNodeOps(NodeOps):
         rescan+enumeration sees it (models hot-add on a box with no fabric)."""
         with self._sim_lock:
             self._sim_detached.discard(device_id)
+        self._invalidate_enum(node)
 
     def refresh_after_detach(self, node: str) -> None:
         for name, hook in self.restart_hooks.items():
