@@ -219,20 +219,7 @@ int cmd_rescan() {
   return 0;
 }
 
-// libcroprobe ABI (cro_amd/hip/probe.hip)
-extern "C" {
-struct CroProbeResult {
-  int ok;
-  int mfma_f32_exact;
-  double hbm_gbps;
-  double bf16_tflops;
-  long long vram_total;
-  long long vram_free;
-  char gcn_arch[64];
-  char msg[256];
-};
-int cro_probe_run(int device, CroProbeResult* out);
-}
+#include "../hip/croprobe.h"
 
 int cmd_probe(int device) {
   CroProbeResult result;

@@ -48,24 +48,7 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 typedef float f32x16 __attribute__((ext_vector_type(16)));
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 
-extern "C" {
-
-struct CroProbeResult {
-  int ok;
-  int mfma_f32_exact;       // 1 = bitwise match vs host fmaf chain
-  double hbm_gbps;          // achieved copy bandwidth (read+write bytes)
-  double bf16_tflops;       // dense bf16 MFMA issue rate
-  long long vram_total;     // bytes
-  long long vram_free;      // bytes
-  double t_setup_ms;        // host wall per section (diagnosing the
-  double t_mfma_ms;         // attach-latency contribution of the probe)
-  double t_bw_ms;
-  double t_bf16_ms;
-  char gcn_arch[64];
-  char msg[256];
-};
-
-}  // extern "C"
+#include "croprobe.h"
 
 // ---------------------------------------------------------------------------
 // 1. exact f32 MFMA check (v_mfma_f32_16x16x4_f32: A 16x4, B 4x16, D 16x16;
