@@ -19,6 +19,8 @@
 #include <dirent.h>
 #include <unistd.h>
 
+#include "../hip/croprobe.h"
+
 #include <cinttypes>
 #include <cstdio>
 #include <cstdlib>
@@ -218,8 +220,6 @@ int cmd_rescan() {
   printf("{\"rescanned\":true}\n");
   return 0;
 }
-
-#include "../hip/croprobe.h"
 
 int cmd_probe(int device) {
   CroProbeResult result;
