@@ -220,6 +220,26 @@ class DeviceTaintRule(K8sObject):
     spec: DeviceTaintRuleSpec = Field(default_factory=DeviceTaintRuleSpec)
 
 
+class DaemonSetSpec(_Model):
+    """Minimal DaemonSet surface for the rolling-restart path: the template
+    annotations carry ``kubectl.kubernetes.io/restartedAt``
+    (nodes.go:35-76)."""
+
+    template_annotations: Dict[str, str] = Field(default_factory=dict)
+
+
+class DaemonSetStatus(_Model):
+    desired_number_scheduled: int = 0
+    number_ready: int = 0
+
+
+class DaemonSet(K8sObject):
+    KIND: ClassVar[str] = "DaemonSet"
+    apiVersion: str = "apps/v1"
+    spec: DaemonSetSpec = Field(default_factory=DaemonSetSpec)
+    status: DaemonSetStatus = Field(default_factory=DaemonSetStatus)
+
+
 ALL_KINDS = {
     cls.KIND: cls
     for cls in (
@@ -228,5 +248,6 @@ ALL_KINDS = {
         Node,
         ResourceSlice,
         DeviceTaintRule,
+        DaemonSet,
     )
 }

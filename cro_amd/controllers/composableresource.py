@@ -104,6 +104,36 @@ class ComposableResourceReconciler(Reconciler):
             self._set_error(resource, str(exc))
             raise
 
+    def _restart_plugin_daemonsets(self, resource: ComposableResource, fatal: bool) -> None:
+        """DEVICE_PLUGIN node refresh: roll the AMD device-plugin and
+        metrics-exporter daemonsets (nvidia-device-plugin-daemonset +
+        nvidia-dcgm parity, composableresource_controller.go:257-269).
+
+        Daemonsets not present in the store mean this deployment does not
+        model them (standalone single-node) — skipped, not an error; a
+        present daemonset that fails to update is surfaced per the
+        reference's attach (log) / detach (fatal) split.
+        """
+        import os
+
+        from ..nodeops.nodes import restart_daemonset
+
+        namespace = os.environ.get("CRO_AMD_GPU_OPERATOR_NAMESPACE", "amd-gpu-operator")
+        for name in ("amd-device-plugin", "amd-metrics-exporter"):
+            try:
+                restart_daemonset(self.client, namespace, name)
+            except NotFoundError:
+                log.debug("daemonset %s/%s not modeled; skipping restart", namespace, name)
+            except Exception as exc:
+                if fatal:
+                    raise
+                log.warning("failed to restart %s/%s: %s", namespace, name, exc)
+                resource.status.error = str(exc)
+                try:
+                    self.client.update_status(resource)
+                except ConflictError:
+                    pass
+
     def _persist_device_identity(
         self, name: str, device_id: str, cdi_device_id: str
     ) -> ComposableResource:
@@ -231,6 +261,7 @@ class ComposableResourceReconciler(Reconciler):
                 self.node_ops.check_no_loads(node)
             except GPULoadsPresent as exc:
                 log.warning("gpu loads during attach on %s: %s", node, exc)
+            self._restart_plugin_daemonsets(resource, fatal=False)
         with self._phase("node_refresh"):
             self.node_ops.refresh_after_attach(node)
 
@@ -323,6 +354,9 @@ class ComposableResourceReconciler(Reconciler):
                     self.adapter.provider.name, "remove"
                 ).observe(time.monotonic() - t0)
 
+            if mode == "DEVICE_PLUGIN":
+                # fatal on detach (composableresource_controller.go:379-385)
+                self._restart_plugin_daemonsets(resource, fatal=True)
             self.node_ops.refresh_after_detach(node)
 
             if mode == "DRA":

@@ -52,6 +52,36 @@ def check_node_capacity_sufficient(
     )
 
 
+RESTARTED_AT_ANNOTATION = "kubectl.kubernetes.io/restartedAt"
+RESTART_DEBOUNCE_SECONDS = 10.0
+
+
+def restart_daemonset(client: Client, namespace: str, name: str) -> bool:
+    """Rolling restart via the ``restartedAt`` template annotation with the
+    10-second debounce guard (nodes.go:35-76 parity).  Returns True when a
+    restart was issued, False when debounced.  Raises NotFoundError when the
+    daemonset does not exist — callers decide whether that is fatal
+    (composableresource_controller.go treats it as an error to surface).
+    """
+    from ..api.v1alpha1.types import DaemonSet
+
+    full_name = f"{namespace}/{name}"
+    ds = client.get(DaemonSet, full_name)
+    last = ds.spec.template_annotations.get(RESTARTED_AT_ANNOTATION, "")
+    if last:
+        try:
+            last_ts = time.mktime(time.strptime(last, "%Y-%m-%dT%H:%M:%SZ"))
+            if time.mktime(time.gmtime()) - last_ts < RESTART_DEBOUNCE_SECONDS:
+                return False  # restarted moments ago — debounce
+        except ValueError:
+            pass  # unparseable stamp → restart anyway
+    ds.spec.template_annotations[RESTARTED_AT_ANNOTATION] = time.strftime(
+        "%Y-%m-%dT%H:%M:%SZ", time.gmtime()
+    )
+    client.update(ds)
+    return True
+
+
 class Debouncer:
     """Per-key debounce: a wrapped call is skipped when the same key fired
     within ``interval`` seconds (nodes.go:56-67 restartedAt guard)."""

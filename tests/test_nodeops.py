@@ -296,3 +296,51 @@ def test_last_gpu_drain_async_error_surfaces():
         ops.drain(NODE, ids[0])
     time.sleep(0.05)
     ops.drain(NODE, ids[0])
+
+
+def test_restart_daemonset_annotation_and_debounce(client):
+    import time as _time
+
+    from cro_amd.api.v1alpha1.types import DaemonSet
+    from cro_amd.nodeops.nodes import (
+        RESTARTED_AT_ANNOTATION,
+        restart_daemonset,
+    )
+    from cro_amd.runtime.errors import NotFoundError
+
+    ds = DaemonSet()
+    ds.metadata.name = "amd-gpu-operator/amd-device-plugin"
+    client.create(ds)
+    assert restart_daemonset(client, "amd-gpu-operator", "amd-device-plugin") is True
+    got = client.get(DaemonSet, "amd-gpu-operator/amd-device-plugin")
+    stamp = got.spec.template_annotations[RESTARTED_AT_ANNOTATION]
+    assert stamp.endswith("Z")
+    # immediate second restart is debounced (nodes.go:56-67)
+    assert restart_daemonset(client, "amd-gpu-operator", "amd-device-plugin") is False
+    got2 = client.get(DaemonSet, "amd-gpu-operator/amd-device-plugin")
+    assert got2.spec.template_annotations[RESTARTED_AT_ANNOTATION] == stamp
+    with pytest.raises(NotFoundError):
+        restart_daemonset(client, "amd-gpu-operator", "nope")
+
+
+def test_device_plugin_mode_restarts_daemonsets(mock_world):
+    from cro_amd.api.v1alpha1.types import ComposableResource, DaemonSet
+    from cro_amd.nodeops.nodes import RESTARTED_AT_ANNOTATION
+
+    mock_world.adapter.device_resource_type = "DEVICE_PLUGIN"
+    for name in ("amd-device-plugin", "amd-metrics-exporter"):
+        ds = DaemonSet()
+        ds.metadata.name = f"amd-gpu-operator/{name}"
+        mock_world.client.create(ds)
+    make_node(mock_world.client, "node0")
+    from tests.conftest import make_resource
+
+    mock_world.client.create(make_resource("gpu-1"))
+    mock_world.resource_rec.reconcile("gpu-1")  # None → Attaching
+    mock_world.resource_rec.reconcile("gpu-1")  # Attaching → Online
+    assert (
+        mock_world.client.get(ComposableResource, "gpu-1").status.state == "Online"
+    )
+    for name in ("amd-device-plugin", "amd-metrics-exporter"):
+        got = mock_world.client.get(DaemonSet, f"amd-gpu-operator/{name}")
+        assert RESTARTED_AT_ANNOTATION in got.spec.template_annotations
