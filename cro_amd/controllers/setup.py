@@ -58,8 +58,9 @@ def build_manager(
     syncer_period: Optional[float] = None,
     syncer_grace: float = 600.0,
     metrics_port: Optional[int] = None,
+    client=None,
 ) -> Manager:
-    mgr = Manager(store=store, metrics_port=metrics_port)
+    mgr = Manager(store=store, metrics_port=metrics_port, client=client)
 
     resource_reconciler = ComposableResourceReconciler(
         mgr.client, adapter, node_ops, resource_config

@@ -21,9 +21,22 @@ from .store import InMemoryStore
 
 
 class Manager:
-    def __init__(self, store: Optional[InMemoryStore] = None, metrics_port: Optional[int] = None):
-        self.store = store or InMemoryStore()
-        self.client = Client(self.store)
+    def __init__(
+        self,
+        store: Optional[InMemoryStore] = None,
+        metrics_port: Optional[int] = None,
+        client=None,
+    ):
+        """``client`` overrides the in-memory store with any object
+        implementing the Client surface plus ``watch``/``list`` (e.g.
+        runtime.remote.RemoteClient) — controllers then run against a remote
+        API server; admission lives server-side in that mode."""
+        if client is not None:
+            self.store = client  # controllers watch/list through it
+            self.client = client
+        else:
+            self.store = store or InMemoryStore()
+            self.client = Client(self.store)
         self.metrics = Metrics()
         self._metrics_port = metrics_port
         self._controllers: List[Controller] = []
@@ -41,6 +54,13 @@ class Manager:
         self._runnables.append((period, fn))
 
     def register_admission(self, kind: str, fn) -> None:
+        if not hasattr(self.store, "register_admission"):
+            import logging
+
+            logging.getLogger(__name__).info(
+                "remote-client mode: admission for %s is enforced server-side", kind
+            )
+            return
         self.store.register_admission(kind, fn)
 
     def start(self) -> None:

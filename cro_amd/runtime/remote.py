@@ -1,0 +1,178 @@
+"""RemoteClient: the Client surface over an apiserver-shaped HTTP API.
+
+Lets the operator process run against a remote API server (the cluster
+deployment shape — controllers on the control plane, node agents on nodes)
+with the same Client/watch surface the in-memory store provides, so every
+controller runs unchanged in either mode.
+
+Watch uses k8s-style list+watch streaming (ndjson); each watched kind gets
+a reader thread feeding the subscriber queue with the same WatchEvent shape
+controllers consume.  Admission and schema validation run server-side.
+"""
+
+from __future__ import annotations
+
+import json
+import logging
+import queue
+import threading
+from typing import Dict, List, Optional, Type, TypeVar, Union
+
+import httpx
+
+from ..api.v1alpha1.types import ALL_KINDS, K8sObject
+from .errors import (
+    AdmissionDenied,
+    AlreadyExistsError,
+    ApiError,
+    ConflictError,
+    NotFoundError,
+)
+from .store import WatchEvent
+
+log = logging.getLogger(__name__)
+
+BASE = "/apis/cro.hpsys.ibm.ie.com/v1alpha1"
+T = TypeVar("T", bound=K8sObject)
+
+_PLURALS = {cls.KIND: plural for plural, cls in {
+    "composabilityrequests": ALL_KINDS["ComposabilityRequest"],
+    "composableresources": ALL_KINDS["ComposableResource"],
+    "resourceslices": ALL_KINDS["ResourceSlice"],
+    "devicetaintrules": ALL_KINDS["DeviceTaintRule"],
+    "nodes": ALL_KINDS["Node"],
+}.items()}
+
+
+def _raise_for(resp: httpx.Response) -> None:
+    if 200 <= resp.status_code < 300:
+        return
+    try:
+        detail = resp.json().get("detail", {})
+        reason = detail.get("reason", "") if isinstance(detail, dict) else ""
+        message = detail.get("message", str(detail)) if isinstance(detail, dict) else str(detail)
+    except ValueError:
+        reason, message = "", resp.text
+    if resp.status_code == 404 or reason == "NotFound":
+        raise NotFoundError(message)
+    if reason == "AlreadyExists":
+        raise AlreadyExistsError(message)
+    if reason == "Conflict" or resp.status_code == 409:
+        raise ConflictError(message)
+    if resp.status_code == 403:
+        raise AdmissionDenied(message)
+    if resp.status_code == 422:
+        raise ValueError(message)
+    raise ApiError(f"{resp.status_code}: {message}")
+
+
+class RemoteClient:
+    def __init__(self, base_url: str, transport: Optional[httpx.BaseTransport] = None):
+        self.base_url = base_url.rstrip("/")
+        self._http = httpx.Client(base_url=self.base_url, transport=transport, timeout=30)
+        self._stop = threading.Event()
+        self._watch_threads: List[threading.Thread] = []
+
+    # -- kind plumbing -----------------------------------------------------
+
+    @staticmethod
+    def _resolve(cls_or_kind: Union[Type[T], str]):
+        if isinstance(cls_or_kind, str):
+            cls = ALL_KINDS[cls_or_kind]
+        else:
+            cls = cls_or_kind
+        return cls, _PLURALS[cls.KIND]
+
+    # -- Client surface ----------------------------------------------------
+
+    def create(self, obj: T) -> T:
+        cls, plural = self._resolve(type(obj))
+        resp = self._http.post(f"{BASE}/{plural}", json=obj.model_dump(by_alias=True))
+        _raise_for(resp)
+        return cls.model_validate(resp.json())
+
+    def get(self, cls_or_kind, name: str):
+        cls, plural = self._resolve(cls_or_kind)
+        resp = self._http.get(f"{BASE}/{plural}/{name}")
+        _raise_for(resp)
+        return cls.model_validate(resp.json())
+
+    def try_get(self, cls_or_kind, name: str):
+        try:
+            return self.get(cls_or_kind, name)
+        except NotFoundError:
+            return None
+
+    def list(self, cls_or_kind, labels: Optional[Dict[str, str]] = None):
+        cls, plural = self._resolve(cls_or_kind)
+        params = {}
+        if labels:
+            params["labelSelector"] = ",".join(f"{k}={v}" for k, v in labels.items())
+        resp = self._http.get(f"{BASE}/{plural}", params=params)
+        _raise_for(resp)
+        return [cls.model_validate(item) for item in resp.json()["items"]]
+
+    def update(self, obj: T) -> T:
+        cls, plural = self._resolve(type(obj))
+        resp = self._http.put(
+            f"{BASE}/{plural}/{obj.metadata.name}", json=obj.model_dump(by_alias=True)
+        )
+        _raise_for(resp)
+        return cls.model_validate(resp.json())
+
+    def update_status(self, obj: T) -> T:
+        cls, plural = self._resolve(type(obj))
+        resp = self._http.put(
+            f"{BASE}/{plural}/{obj.metadata.name}/status",
+            json=obj.model_dump(by_alias=True),
+        )
+        _raise_for(resp)
+        return cls.model_validate(resp.json())
+
+    def delete(self, obj_or_cls, name: Optional[str] = None) -> None:
+        if name is None:
+            cls, plural = self._resolve(type(obj_or_cls))
+            name = obj_or_cls.metadata.name
+        else:
+            cls, plural = self._resolve(obj_or_cls)
+        resp = self._http.delete(f"{BASE}/{plural}/{name}")
+        _raise_for(resp)
+
+    # -- watch (Controller.start_watch consumes this) ----------------------
+
+    def watch(self, kinds: Optional[List[str]] = None) -> "queue.Queue[WatchEvent]":
+        q: "queue.Queue[WatchEvent]" = queue.Queue()
+        for kind in kinds or list(_PLURALS):
+            t = threading.Thread(
+                target=self._watch_kind, args=(kind, q),
+                name=f"remote-watch-{kind}", daemon=True,
+            )
+            t.start()
+            self._watch_threads.append(t)
+        return q
+
+    def _watch_kind(self, kind: str, q: "queue.Queue[WatchEvent]") -> None:
+        cls, plural = self._resolve(kind)
+        # the stream's initial list doubles as the informer-cache replay;
+        # reconnects re-list (level-triggered reconciles tolerate replays)
+        while not self._stop.is_set():
+            try:
+                with self._http.stream(
+                    "GET", f"{BASE}/{plural}", params={"watch": "true"}, timeout=None
+                ) as resp:
+                    for line in resp.iter_lines():
+                        if self._stop.is_set():
+                            return
+                        if not line.strip():
+                            continue  # keepalive
+                        ev = json.loads(line)
+                        q.put(WatchEvent(ev["type"], cls.model_validate(ev["object"])))
+            except Exception as exc:
+                if self._stop.is_set():
+                    return
+                log.debug("watch %s disconnected (%s); reconnecting", kind, exc)
+                self._stop.wait(0.2)
+
+    def close(self) -> None:
+        self._stop.set()
+        self._http.close()

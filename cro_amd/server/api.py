@@ -58,17 +58,20 @@ def _dump(obj) -> dict:
 
 
 def _http_error(exc: Exception) -> HTTPException:
+    """Error taxonomy over HTTP: the reason field lets a remote client map
+    back to the exact error class (409 covers both AlreadyExists and
+    optimistic-concurrency Conflict, as in the k8s API)."""
     if isinstance(exc, NotFoundError):
-        return HTTPException(404, str(exc))
+        return HTTPException(404, {"reason": "NotFound", "message": str(exc)})
     if isinstance(exc, AlreadyExistsError):
-        return HTTPException(409, str(exc))
+        return HTTPException(409, {"reason": "AlreadyExists", "message": str(exc)})
     if isinstance(exc, ConflictError):
-        return HTTPException(409, str(exc))
+        return HTTPException(409, {"reason": "Conflict", "message": str(exc)})
     if isinstance(exc, AdmissionDenied):
-        return HTTPException(403, str(exc))
+        return HTTPException(403, {"reason": "Forbidden", "message": str(exc)})
     if isinstance(exc, ValueError):
-        return HTTPException(422, str(exc))
-    return HTTPException(500, str(exc))
+        return HTTPException(422, {"reason": "Invalid", "message": str(exc)})
+    return HTTPException(500, {"reason": "InternalError", "message": str(exc)})
 
 
 def build_app(client: Client) -> FastAPI:
@@ -92,8 +95,10 @@ def build_app(client: Client) -> FastAPI:
         )
 
     @app.get(BASE + "/{plural}")
-    def list_objects(plural: str, labelSelector: str = ""):
+    async def list_objects(plural: str, labelSelector: str = "", watch: bool = False):
         cls = _cls(plural)
+        if watch:
+            return await _watch_stream(cls)
         labels = None
         if labelSelector:
             labels = dict(part.split("=", 1) for part in labelSelector.split(","))
@@ -103,6 +108,30 @@ def build_app(client: Client) -> FastAPI:
             "kind": cls.KIND + "List",
             "items": [_dump(o) for o in items],
         }
+
+    async def _watch_stream(cls):
+        """k8s-style list+watch: current objects as synthetic ADDED events,
+        then live events, newline-delimited JSON."""
+        import asyncio
+        import json as _json
+        import queue as _queue
+
+        from fastapi.responses import StreamingResponse
+
+        async def gen():
+            events = client.watch([cls.KIND])
+            for obj in client.list(cls):
+                yield _json.dumps({"type": "ADDED", "object": _dump(obj)}) + "\n"
+            loop = asyncio.get_running_loop()
+            while True:
+                try:
+                    ev = await loop.run_in_executor(None, events.get, True, 1.0)
+                except _queue.Empty:
+                    yield "\n"  # keepalive; also surfaces disconnects
+                    continue
+                yield _json.dumps({"type": ev.type, "object": _dump(ev.object)}) + "\n"
+
+        return StreamingResponse(gen(), media_type="application/x-ndjson")
 
     @app.get(BASE + "/{plural}/{name:path}")
     def get_object(plural: str, name: str):

@@ -51,7 +51,7 @@ def test_admission_enforced_over_http(api_stack):
     bad = make_request("r1", policy="differentnode", target_node="node0")
     resp = http.post(PLURAL, json=bad.model_dump(by_alias=True))
     assert resp.status_code == 403
-    assert "TargetNode cannot be specified" in resp.json()["detail"]
+    assert "TargetNode cannot be specified" in resp.json()["detail"]["message"]
 
 
 def test_schema_validation_over_http(api_stack):

@@ -1,0 +1,139 @@
+"""Distributed deployment shape: an API-server process (uvicorn over the
+in-memory store, admission server-side) driven by a SEPARATE operator
+manager running all controllers through RemoteClient watch streams —
+the cluster topology, exercised over real sockets."""
+
+import socket
+import threading
+import time
+
+import pytest
+
+from cro_amd.api.v1alpha1.types import ComposabilityRequest, ComposableResource, Node
+from cro_amd.controllers import build_manager
+from cro_amd.fabric.adapter import Adapter
+from cro_amd.fabric.mock import MockFabric
+from cro_amd.nodeops.amdgpu import MockNodeOps
+from cro_amd.runtime.errors import AdmissionDenied, NotFoundError
+from cro_amd.runtime.manager import Manager
+from cro_amd.runtime.remote import RemoteClient
+from cro_amd.server.api import build_app
+from tests.conftest import make_request
+
+
+def free_port() -> int:
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+@pytest.fixture
+def api_server():
+    """uvicorn serving the store + admission — no controllers."""
+    import uvicorn
+
+    server_mgr = build_manager(Adapter("DRA", MockFabric()), None)  # store+admission only
+    port = free_port()
+    app = build_app(server_mgr.client)
+    server = uvicorn.Server(
+        uvicorn.Config(app, host="127.0.0.1", port=port, log_level="error")
+    )
+    thread = threading.Thread(target=server.run, daemon=True)
+    thread.start()
+    deadline = time.monotonic() + 15
+    import httpx
+
+    while time.monotonic() < deadline:
+        try:
+            if httpx.get(f"http://127.0.0.1:{port}/healthz", timeout=1).status_code == 200:
+                break
+        except Exception:
+            time.sleep(0.05)
+    else:
+        pytest.fail("api server did not come up")
+    yield f"http://127.0.0.1:{port}", server_mgr
+    server.should_exit = True
+    thread.join(timeout=5)
+
+
+@pytest.fixture
+def remote_operator(api_server):
+    """A full operator manager whose only link to the world is HTTP."""
+    url, server_mgr = api_server
+    remote = RemoteClient(url)
+    fabric = MockFabric(models={"mi355x": 8})
+    mgr = build_manager(
+        Adapter("DRA", fabric), None, client=remote, enable_webhook=False
+    )
+    ops = MockNodeOps(client=mgr.client)
+    mgr.resource_reconciler.node_ops = ops
+
+    orig_add = fabric.add_resource
+
+    def add_resource(resource):
+        did, cdi = orig_add(resource)
+        ops.fabric_composed(resource.spec.target_node, did)
+        return did, cdi
+
+    fabric.add_resource = add_resource
+
+    node = Node()
+    node.metadata.name = "node0"
+    remote.create(node)
+    ops.set_driver("node0", True)
+    mgr.start()
+    yield mgr, remote, fabric, server_mgr
+    mgr.stop()
+    remote.close()
+
+
+def wait_for(predicate, timeout=20.0):
+    deadline = time.monotonic() + timeout
+    while time.monotonic() < deadline:
+        if predicate():
+            return True
+        time.sleep(0.02)
+    return predicate()
+
+
+def test_remote_crud_and_errors(api_server):
+    url, _ = api_server
+    remote = RemoteClient(url)
+    try:
+        created = remote.create(make_request("r1", target_node="node0"))
+        assert created.metadata.uid
+        assert remote.get(ComposabilityRequest, "r1").metadata.name == "r1"
+        with pytest.raises(NotFoundError):
+            remote.get(ComposabilityRequest, "ghost")
+        # admission enforced server-side
+        with pytest.raises(AdmissionDenied):
+            remote.create(make_request("dup", target_node="node0"))
+        created.spec.resource.size = 2
+        updated = remote.update(created)
+        assert updated.spec.resource.size == 2
+        updated.status.state = "NodeAllocating"
+        assert remote.update_status(updated).status.state == "NodeAllocating"
+        remote.delete(ComposabilityRequest, "r1")
+        assert remote.try_get(ComposabilityRequest, "r1") is None
+    finally:
+        remote.close()
+
+
+def test_remote_operator_full_lifecycle(remote_operator):
+    mgr, remote, fabric, server_mgr = remote_operator
+    remote.create(make_request("r1", size=2, target_node="node0"))
+    assert wait_for(
+        lambda: (req := remote.try_get(ComposabilityRequest, "r1")) is not None
+        and req.status.state == "Running"
+    ), (remote.try_get(ComposabilityRequest, "r1") or object()).__dict__
+    req = remote.get(ComposabilityRequest, "r1")
+    assert len(req.status.resources) == 2
+    assert all(v.state == "Online" for v in req.status.resources.values())
+    assert len(fabric.attached_to("node0")) == 2
+
+    remote.delete(ComposabilityRequest, "r1")
+    assert wait_for(lambda: remote.try_get(ComposabilityRequest, "r1") is None)
+    assert wait_for(lambda: fabric.attached_to("node0") == [])
+    assert server_mgr.client.list(ComposableResource) == []
