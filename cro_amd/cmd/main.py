@@ -47,6 +47,12 @@ def main(argv=None) -> int:
     p.add_argument("--leader-elect-lock", default="/var/run/cro-amd/leader.lock")
     p.add_argument("--node", default=os.environ.get("NODE_NAME", ""),
                    help="local node name this operator instance manages")
+    p.add_argument("--api-server", default="",
+                   help="run controllers against a remote cro-amd API server "
+                   "(URL) instead of the embedded store")
+    p.add_argument("--serve-only", action="store_true",
+                   help="serve the API/store without running controllers "
+                   "(the apiserver half of a split deployment)")
     p.add_argument("--cdi-dir", default="/etc/cdi")
     p.add_argument("--destructive", action="store_true",
                    help="perform real PCI remove/rescan and module unload")
@@ -76,9 +82,15 @@ def main(argv=None) -> int:
 
     adapter = new_adapter()
 
+    remote = None
+    if args.api_server:
+        from ..runtime.remote import RemoteClient
+
+        remote = RemoteClient(args.api_server)
     mgr = build_manager(
         adapter,
         None,
+        client=remote,
         max_concurrent_reconciles=args.max_concurrent_reconciles,
         enable_webhook=os.environ.get("ENABLE_WEBHOOKS", "") != "false",
         syncer_period=args.syncer_period,
@@ -103,9 +115,12 @@ def main(argv=None) -> int:
     if hasattr(mgr, "syncer"):
         mgr.syncer.node_ops = node_ops
 
-    mgr.start()
-    log.info("manager started (%d reconcile workers per controller)",
-             args.max_concurrent_reconciles)
+    if not args.serve_only:
+        mgr.start()
+        log.info("manager started (%d reconcile workers per controller)",
+                 args.max_concurrent_reconciles)
+    else:
+        log.info("serve-only mode: API/store up, controllers disabled")
 
     from ..server.api import build_app
 
