@@ -18,8 +18,6 @@ from typing import ClassVar, Dict, List, Optional
 
 from pydantic import BaseModel, ConfigDict, Field
 
-from .. import _schema_validation as sv  # noqa: F401  (re-exported helpers)
-
 GROUP = "cro.hpsys.ibm.ie.com"
 VERSION = "v1alpha1"
 API_VERSION = f"{GROUP}/{VERSION}"

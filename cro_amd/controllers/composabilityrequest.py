@@ -36,7 +36,7 @@ from ..nodeops.nodes import (
 )
 from ..runtime.client import Client
 from ..runtime.controller import Reconciler, Result
-from ..runtime.errors import ConflictError, NotFoundError
+from ..runtime.errors import ConflictError
 from ..utils import generate_composable_resource_name
 from .composableresource import FINALIZER, MANAGED_BY_LABEL, READY_TO_DETACH_LABEL
 

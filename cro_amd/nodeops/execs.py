@@ -20,7 +20,7 @@ from __future__ import annotations
 import os
 import subprocess
 import threading
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 TRUSTED_BIN_DIRS = (
     "/opt/rocm/bin",

@@ -38,7 +38,7 @@ from typing import Callable, Dict, List, Optional
 
 from ..api import _schema_validation
 from ..api.v1alpha1.types import K8sObject
-from .errors import AdmissionDenied, AlreadyExistsError, ConflictError, NotFoundError
+from .errors import AlreadyExistsError, ConflictError, NotFoundError
 
 ADDED = "ADDED"
 MODIFIED = "MODIFIED"
