@@ -1,0 +1,76 @@
+"""Node-agent HTTP API: remote node access for off-node controllers.
+
+The reference reaches node hardware by SPDY-exec'ing shell pipelines into
+privileged pods (gpus.go:1040-1067).  Here the node agent is a first-class
+process; controllers running off-node reach its NodeExec surface over this
+API (cro_amd/runtime exposes the matching client, RemoteNodeExec):
+
+    POST /agent/run        {"argv": [...], "timeout": s} → {rc, stdout, stderr}
+    GET  /agent/file?path= → raw contents          (404 on missing)
+    PUT  /agent/file?path= (body = contents)
+    GET  /agent/dir?path=  → {"entries": [...]}
+    GET  /agent/exists?path= → {"exists": bool}
+
+Only the NodeExec verbs are exposed — no arbitrary shell.  Binary execution
+inherits LocalNodeExec's trusted-path resolution (never $PATH).
+"""
+
+from __future__ import annotations
+
+from fastapi import FastAPI, HTTPException, Request, Response
+
+from ..nodeops.execs import ExecError, NodeExec
+
+
+def build_agent_app(execer: NodeExec, node_name: str = "local") -> FastAPI:
+    app = FastAPI(title="cro-amd node agent")
+
+    @app.get("/healthz")
+    def healthz():
+        return {"status": "ok", "node": node_name}
+
+    @app.post("/agent/run")
+    async def run(request: Request):
+        body = await request.json()
+        argv = body.get("argv", [])
+        if not argv or not isinstance(argv, list):
+            raise HTTPException(422, "argv must be a non-empty list")
+        timeout = float(body.get("timeout", 60.0))
+        try:
+            rc, out, err = execer.run(node_name, [str(a) for a in argv], timeout=timeout)
+        except ExecError as exc:
+            raise HTTPException(400, str(exc))
+        return {"rc": rc, "stdout": out, "stderr": err}
+
+    @app.get("/agent/file")
+    def read_file(path: str):
+        try:
+            return Response(execer.read_file(node_name, path), media_type="text/plain")
+        except FileNotFoundError:
+            raise HTTPException(404, f"{path} not found")
+        except (PermissionError, OSError) as exc:
+            raise HTTPException(403, str(exc))
+
+    @app.put("/agent/file")
+    async def write_file(path: str, request: Request):
+        data = (await request.body()).decode()
+        try:
+            execer.write_file(node_name, path, data)
+        except (PermissionError, OSError) as exc:
+            raise HTTPException(403, str(exc))
+        return {"written": path}
+
+    @app.get("/agent/dir")
+    def list_dir(path: str):
+        try:
+            return {"entries": execer.list_dir(node_name, path)}
+        except FileNotFoundError:
+            raise HTTPException(404, f"{path} not found")
+        except (PermissionError, OSError) as exc:
+            raise HTTPException(403, str(exc))
+
+    @app.get("/agent/exists")
+    def exists(path: str):
+        return {"exists": execer.path_exists(node_name, path)}
+
+    return app
