@@ -123,10 +123,13 @@ def main(argv=None) -> int:
         log.info("serve-only mode: API/store up, controllers disabled")
 
     from ..server.api import build_app
+    from ..server.agent_api import build_agent_app
 
     import uvicorn
 
     app = build_app(mgr.client)
+    if args.node:  # node-agent surface for off-node controllers
+        build_agent_app(node_ops.execer, node_name=args.node, app=app)
     server = uvicorn.Server(
         uvicorn.Config(
             app,

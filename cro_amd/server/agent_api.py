@@ -22,12 +22,18 @@ from fastapi import FastAPI, HTTPException, Request, Response
 from ..nodeops.execs import ExecError, NodeExec
 
 
-def build_agent_app(execer: NodeExec, node_name: str = "local") -> FastAPI:
-    app = FastAPI(title="cro-amd node agent")
+def build_agent_app(
+    execer: NodeExec, node_name: str = "local", app: FastAPI = None
+) -> FastAPI:
+    """Build the agent app, or graft the /agent routes onto an existing app
+    (the operator entrypoint serves API + agent surface in one process)."""
+    standalone = app is None
+    if standalone:
+        app = FastAPI(title="cro-amd node agent")
 
-    @app.get("/healthz")
-    def healthz():
-        return {"status": "ok", "node": node_name}
+        @app.get("/healthz")
+        def healthz():
+            return {"status": "ok", "node": node_name}
 
     @app.post("/agent/run")
     async def run(request: Request):
