@@ -141,3 +141,101 @@ def test_bench_one_gpu_quick():
     result = json.loads(proc.stdout.strip().splitlines()[-1])
     assert result["config"]["node_path"] == "real KFD/CDI/HIP-probe"
     assert result["value"] < 30000  # ms; beat the reference's poll quantum
+
+
+def test_split_deployment_real_node_path():
+    """Cluster topology on real hardware: uvicorn API server (store +
+    admission), operator over RemoteClient watch streams, REAL amdgpu node
+    path (KFD/CDI/probe)."""
+    _require_gpu()
+    import socket
+    import threading
+    import time
+
+    import httpx
+    import uvicorn
+
+    from cro_amd.api.v1alpha1.types import ComposabilityRequest, Node
+    from cro_amd.controllers import build_manager
+    from cro_amd.fabric.adapter import Adapter
+    from cro_amd.fabric.mock import MockFabric
+    from cro_amd.nodeops.amdgpu import AmdNodeOps
+    from cro_amd.nodeops.execs import LocalNodeExec
+    from cro_amd.nodeops.kfd import enumerate_gpus
+    from cro_amd.runtime.remote import RemoteClient
+    from cro_amd.server.api import build_app
+    from tests.conftest import make_request
+
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+
+    server_mgr = build_manager(Adapter("DRA", MockFabric()), None)
+    server = uvicorn.Server(
+        uvicorn.Config(build_app(server_mgr.client), host="127.0.0.1", port=port,
+                       log_level="error")
+    )
+    thread = threading.Thread(target=server.run, daemon=True)
+    thread.start()
+    deadline = time.monotonic() + 15
+    while time.monotonic() < deadline:
+        try:
+            if httpx.get(f"http://127.0.0.1:{port}/healthz", timeout=1).status_code == 200:
+                break
+        except Exception:
+            time.sleep(0.05)
+
+    execer = LocalNodeExec()
+    gpus = enumerate_gpus(execer, "gpunode")[:1]
+    fabric = MockFabric(bind_inventory=[
+        {"device_id": g.device_id, "cdi_device_id": f"amd.com/gpu={g.device_id}",
+         "model": "mi355x"} for g in gpus
+    ])
+    remote = RemoteClient(f"http://127.0.0.1:{port}")
+    mgr = build_manager(Adapter("DRA", fabric), None, client=remote, enable_webhook=False)
+    cdi_dir = os.path.join(os.environ.get("TMPDIR", "/tmp"), "cro-cdi-split")
+    ops = AmdNodeOps(
+        execer, client=mgr.client, cdi_dir=cdi_dir, destructive=False,
+        initially_detached=[g.device_id for g in gpus],
+    )
+    mgr.resource_reconciler.node_ops = ops
+    orig_add = fabric.add_resource
+
+    def add_resource(resource):
+        did, cdi = orig_add(resource)
+        ops.simulate_compose(resource.spec.target_node, did)
+        return did, cdi
+
+    fabric.add_resource = add_resource
+    node = Node()
+    node.metadata.name = "gpunode"
+    remote.create(node)
+    mgr.start()
+    try:
+        remote.create(make_request("split-r1", size=1, target_node="gpunode"))
+        deadline = time.monotonic() + 60
+        while time.monotonic() < deadline:
+            req = remote.try_get(ComposabilityRequest, "split-r1")
+            if req is not None and req.status.state == "Running":
+                break
+            time.sleep(0.05)
+        req = remote.get(ComposabilityRequest, "split-r1")
+        assert req.status.state == "Running", req.status
+        device_id = next(iter(req.status.resources.values())).device_id
+        assert device_id == gpus[0].device_id  # the real GPU, over HTTP
+        assert ops.cdi.devices("gpunode") == [device_id]
+
+        remote.delete(ComposabilityRequest, "split-r1")
+        deadline = time.monotonic() + 60
+        while time.monotonic() < deadline:
+            if remote.try_get(ComposabilityRequest, "split-r1") is None:
+                break
+            time.sleep(0.05)
+        assert remote.try_get(ComposabilityRequest, "split-r1") is None
+        assert ops.cdi.devices("gpunode") == []
+    finally:
+        mgr.stop()
+        remote.close()
+        server.should_exit = True
+        thread.join(timeout=5)
