@@ -239,3 +239,43 @@ def test_split_deployment_real_node_path():
         remote.close()
         server.should_exit = True
         thread.join(timeout=5)
+
+
+def test_churn_races_on_real_node_path():
+    """Abort-mid-attach races against the REAL node path: requests deleted
+    at random points of their lifecycle must never leak CDI entries or
+    fabric attachments (the syncer heals the crash-window leaks)."""
+    _require_gpu()
+    import random
+    import time
+
+    from cro_amd.api.v1alpha1.types import ComposabilityRequest
+    from cro_amd.bench_harness import build_local_stack
+    from tests.conftest import make_request
+
+    stack = build_local_stack(
+        node_name="gpuchurn", use_gpu=True, gpu_index=0,
+        cdi_dir=os.path.join(os.environ.get("TMPDIR", "/tmp"), "cro-cdi-churn"),
+        syncer_period=0.5, syncer_grace=2.0,
+    )
+    stack.mgr.start()
+    rng = random.Random(7)
+    try:
+        for i in range(25):
+            name = f"churn-{i}"
+            req = make_request(name, size=1, target_node="gpuchurn")
+            stack.mgr.client.create(req)
+            time.sleep(rng.random() * 0.02)  # abort at a random point
+            stack.mgr.client.delete(ComposabilityRequest, name)
+            assert stack.mgr.wait_for(
+                lambda: stack.mgr.client.try_get(ComposabilityRequest, name) is None,
+                timeout=30,
+            ), f"cycle {i} wedged"
+        # converged: nothing attached, CDI clean (allow the syncer a moment
+        # for any crash-window leak)
+        assert stack.mgr.wait_for(
+            lambda: stack.fabric.attached_to("gpuchurn") == [], timeout=20
+        )
+        assert stack.ops.cdi.devices("gpuchurn") == []
+    finally:
+        stack.mgr.stop()

@@ -66,3 +66,14 @@ def test_self_update_not_a_conflict(guarded_client):
     created = guarded_client.create(make_request("r1", target_node="node0"))
     created.spec.resource.size = 4
     guarded_client.update(created)  # same name — excluded from conflict scan
+
+
+def test_two_unpinned_samenode_requests_conflict(guarded_client):
+    """Two samenode requests with no target node and the same (type, model)
+    collide on the empty implicit target — rejected, matching the
+    reference's comparison semantics (webhook :107-128: "" == "")."""
+    guarded_client.create(make_request("r1"))
+    with pytest.raises(AdmissionDenied, match="already exists"):
+        guarded_client.create(make_request("r2"))
+    # a different model is fine
+    guarded_client.create(make_request("r3", model="mi300x"))
