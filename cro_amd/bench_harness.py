@@ -157,10 +157,15 @@ def build_local_stack(
     node.status.allocatable.allowed_pod_number = 256
     mgr.client.create(node)
 
-    return LocalStack(
+    stack = LocalStack(
         mgr=mgr, fabric=fabric, ops=ops, node_name=node_name,
         device_ids=device_ids, gpu=use_gpu,
     )
+    # one long-lived request watch for event-driven waits in
+    # attach_detach_cycle (polling would add its granularity to the
+    # measured latency)
+    stack.request_events = mgr.store.watch(["ComposabilityRequest"])
+    return stack
 
 
 def attach_detach_cycle(stack: LocalStack, name: str, size: int = 1, timeout: float = 60.0, force_detach: bool = False) -> dict:
@@ -169,6 +174,8 @@ def attach_detach_cycle(stack: LocalStack, name: str, size: int = 1, timeout: fl
     attach_ms = create → Running (every device Online with CDI written);
     detach_ms = delete → object gone (device drained + fabric detach done).
     """
+    import queue as _queue
+
     mgr = stack.mgr
     req = ComposabilityRequest(
         spec=ComposabilityRequestSpec(
@@ -180,13 +187,36 @@ def attach_detach_cycle(stack: LocalStack, name: str, size: int = 1, timeout: fl
     )
     req.metadata.name = name
 
+    events = getattr(stack, "request_events", None)
+
+    def wait_event(pred) -> bool:
+        """Event-driven wait on the stack's request watch (no poll
+        granularity in the measured latency); falls back to polling when
+        no watch is attached."""
+        deadline = time.monotonic() + timeout
+        while True:
+            remaining = deadline - time.monotonic()
+            if remaining <= 0:
+                return False
+            try:
+                ev = events.get(timeout=remaining)
+            except _queue.Empty:
+                return False
+            if ev.object.metadata.name == name and pred(ev):
+                return True
+
     t0 = time.monotonic()
     mgr.client.create(req)
-    ok = mgr.wait_for(
-        lambda: mgr.client.try_get(ComposabilityRequest, name) is not None
-        and mgr.client.get(ComposabilityRequest, name).status.state == "Running",
-        timeout=timeout,
-    )
+    if events is not None:
+        ok = wait_event(
+            lambda ev: ev.type != "DELETED" and ev.object.status.state == "Running"
+        )
+    else:
+        ok = mgr.wait_for(
+            lambda: mgr.client.try_get(ComposabilityRequest, name) is not None
+            and mgr.client.get(ComposabilityRequest, name).status.state == "Running",
+            timeout=timeout,
+        )
     t1 = time.monotonic()
     if not ok:
         cur = mgr.client.try_get(ComposabilityRequest, name)
@@ -197,9 +227,13 @@ def attach_detach_cycle(stack: LocalStack, name: str, size: int = 1, timeout: fl
         )
 
     mgr.client.delete(ComposabilityRequest, name)
-    ok = mgr.wait_for(
-        lambda: mgr.client.try_get(ComposabilityRequest, name) is None, timeout=timeout
-    )
+    if events is not None:
+        ok = wait_event(lambda ev: ev.type == "DELETED")
+    else:
+        ok = mgr.wait_for(
+            lambda: mgr.client.try_get(ComposabilityRequest, name) is None,
+            timeout=timeout,
+        )
     t2 = time.monotonic()
     if not ok:
         from .api.v1alpha1.types import ComposableResource

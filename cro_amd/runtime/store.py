@@ -97,6 +97,15 @@ class InMemoryStore:
             self._watchers.append(w)
         return w.queue
 
+    def stop_watch(self, q: "queue.Queue[WatchEvent]") -> None:
+        """Unsubscribe a watcher (long-lived processes creating transient
+        watches must release them or every event fans out to dead queues)."""
+        with self._lock:
+            for w in self._watchers:
+                if w.queue is q:
+                    w.closed = True
+            self._watchers = [w for w in self._watchers if not w.closed]
+
     def _notify(self, ev: WatchEvent) -> None:
         # deep copies per subscriber so no watcher can mutate shared state
         for w in self._watchers:
