@@ -173,7 +173,16 @@ def enumerate_gpus(execer: NodeExec, node: str) -> List[GPUDevice]:
 
 
 def enumerate_gpus_amdsmi(execer: NodeExec, node: str) -> List[GPUDevice]:
-    """Fallback enumeration via ``amd-smi list --json`` (CLI fork — slow path)."""
+    """Fallback enumeration via ``amd-smi list --json`` (CLI fork — slow path).
+
+    Output rows look like ``{"gpu": 0, "bdf": "0000:f1:00.0", "uuid":
+    "d0ff75a3-…", "kfd_id": 8465, "node_id": 6, "partition_id": 0}``
+    (captured on MI355X/ROCm 7.2).  Caveats vs the KFD path: the amd-smi
+    uuid is NOT the KFD unique_id fuse, so device ids from this path only
+    match each other (use one enumeration source consistently), and no
+    render minor is reported — CDI emission needs the KFD path.
+    ``kfd_id`` is the KFD gpu_id used for /sys/class/kfd/kfd/proc load
+    attribution."""
     rc, out, err = execer.run(node, ["amd-smi", "list", "--json"])
     if rc != 0:
         raise ExecError(f"amd-smi list failed: {err}", rc=rc, stderr=err)
@@ -191,8 +200,8 @@ def enumerate_gpus_amdsmi(execer: NodeExec, node: str) -> List[GPUDevice]:
             uid = 0
         gpus.append(
             GPUDevice(
-                kfd_node=-1,
-                gpu_id=row.get("gpu", -1),
+                kfd_node=row.get("node_id", -1),
+                gpu_id=row.get("kfd_id", -1),
                 device_id=canonical_device_id(uid, bdf),
                 unique_id=uid,
                 render_minor=0,

@@ -344,3 +344,24 @@ def test_device_plugin_mode_restarts_daemonsets(mock_world):
     for name in ("amd-device-plugin", "amd-metrics-exporter"):
         got = mock_world.client.get(DaemonSet, f"amd-gpu-operator/{name}")
         assert RESTARTED_AT_ANNOTATION in got.spec.template_annotations
+
+
+def test_amdsmi_fallback_parses_real_format():
+    """Parser vs a captured real `amd-smi list --json` payload
+    (MI355X / ROCm 7.2)."""
+    from cro_amd.nodeops.kfd import enumerate_gpus_amdsmi
+
+    ex = MockNodeExec()
+    ex.set_command(
+        ("amd-smi", "list", "--json"),
+        (0, '[{"gpu": 0, "bdf": "0000:f1:00.0", '
+            '"uuid": "d0ff75a3-0000-1000-805e-ceb808463c96", '
+            '"kfd_id": 8465, "node_id": 6, "partition_id": 0}]', ""),
+    )
+    gpus = enumerate_gpus_amdsmi(ex, NODE)
+    assert len(gpus) == 1
+    g = gpus[0]
+    assert g.pci_bdf == "0000:f1:00.0"
+    assert g.gpu_id == 8465  # kfd_id drives /sys/class/kfd/kfd/proc attribution
+    assert g.kfd_node == 6
+    assert g.device_id.startswith("GPU-")
