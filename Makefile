@@ -27,6 +27,7 @@ manifests:
 
 lint:
 	$(PYTHON) -m compileall -q cro_amd tests bench.py __graft_entry__.py
+	$(PYTHON) tools/lint_imports.py
 
 docker-build:
 	docker build -t cro-amd-operator:latest .
