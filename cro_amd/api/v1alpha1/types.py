@@ -229,6 +229,9 @@ class DaemonSetSpec(_Model):
 class DaemonSetStatus(_Model):
     desired_number_scheduled: int = 0
     number_ready: int = 0
+    current_number_scheduled: int = 0
+    number_unavailable: int = 0
+    number_misscheduled: int = 0
 
 
 class DaemonSet(K8sObject):

@@ -58,23 +58,39 @@ RESTART_DEBOUNCE_SECONDS = 10.0
 
 def restart_daemonset(client: Client, namespace: str, name: str) -> bool:
     """Rolling restart via the ``restartedAt`` template annotation with the
-    10-second debounce guard (nodes.go:35-76 parity).  Returns True when a
-    restart was issued, False when debounced.  Raises NotFoundError when the
-    daemonset does not exist — callers decide whether that is fatal
-    (composableresource_controller.go treats it as an error to surface).
+    stability + 10-second debounce guards (nodes.go:35-76 parity): skipped
+    when the daemonset has nothing scheduled or a rollout is already in
+    flight (ready < desired, current < desired, unavailable, misscheduled),
+    or when it restarted within the debounce window.  Returns True when a
+    restart was issued.  Raises NotFoundError when the daemonset does not
+    exist and ValueError on an unparseable restartedAt stamp (reference
+    errors there too).
     """
     from ..api.v1alpha1.types import DaemonSet
 
     full_name = f"{namespace}/{name}"
     ds = client.get(DaemonSet, full_name)
+    st = ds.status
+    if st.desired_number_scheduled == 0:
+        return False  # nothing scheduled — restart would be a no-op
+    if (
+        st.number_ready < st.desired_number_scheduled
+        or st.current_number_scheduled < st.desired_number_scheduled
+        or st.number_unavailable > 0
+        or st.number_misscheduled > 0
+    ):
+        return False  # rollout in flight / unstable — do not stack restarts
     last = ds.spec.template_annotations.get(RESTARTED_AT_ANNOTATION, "")
     if last:
         try:
             last_ts = time.mktime(time.strptime(last, "%Y-%m-%dT%H:%M:%SZ"))
-            if time.mktime(time.gmtime()) - last_ts < RESTART_DEBOUNCE_SECONDS:
-                return False  # restarted moments ago — debounce
-        except ValueError:
-            pass  # unparseable stamp → restart anyway
+        except ValueError as exc:
+            raise ValueError(
+                f"failed to parse restartedAt annotation for DaemonSet "
+                f"{full_name}: {exc}"
+            )
+        if time.mktime(time.gmtime()) - last_ts < RESTART_DEBOUNCE_SECONDS:
+            return False  # restarted moments ago — debounce
     ds.spec.template_annotations[RESTARTED_AT_ANNOTATION] = time.strftime(
         "%Y-%m-%dT%H:%M:%SZ", time.gmtime()
     )

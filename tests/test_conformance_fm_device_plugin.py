@@ -77,6 +77,9 @@ def dp_stack():
     for name in ("amd-device-plugin", "amd-metrics-exporter"):
         ds = DaemonSet()
         ds.metadata.name = f"amd-gpu-operator/{name}"
+        ds.status.desired_number_scheduled = 1
+        ds.status.number_ready = 1
+        ds.status.current_number_scheduled = 1
         mgr.client.create(ds)
 
     orig_add = provider.add_resource

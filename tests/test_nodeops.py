@@ -310,6 +310,9 @@ def test_restart_daemonset_annotation_and_debounce(client):
 
     ds = DaemonSet()
     ds.metadata.name = "amd-gpu-operator/amd-device-plugin"
+    ds.status.desired_number_scheduled = 1
+    ds.status.number_ready = 1
+    ds.status.current_number_scheduled = 1
     client.create(ds)
     assert restart_daemonset(client, "amd-gpu-operator", "amd-device-plugin") is True
     got = client.get(DaemonSet, "amd-gpu-operator/amd-device-plugin")
@@ -331,6 +334,9 @@ def test_device_plugin_mode_restarts_daemonsets(mock_world):
     for name in ("amd-device-plugin", "amd-metrics-exporter"):
         ds = DaemonSet()
         ds.metadata.name = f"amd-gpu-operator/{name}"
+        ds.status.desired_number_scheduled = 1
+        ds.status.number_ready = 1
+        ds.status.current_number_scheduled = 1
         mock_world.client.create(ds)
     make_node(mock_world.client, "node0")
     from tests.conftest import make_resource
@@ -365,3 +371,46 @@ def test_amdsmi_fallback_parses_real_format():
     assert g.gpu_id == 8465  # kfd_id drives /sys/class/kfd/kfd/proc attribution
     assert g.kfd_node == 6
     assert g.device_id.startswith("GPU-")
+
+
+def test_restart_daemonset_stability_guards(client):
+    """Stability guards (nodes.go:40-50): no restart when nothing is
+    scheduled or a rollout is in flight; unparseable stamp errors."""
+    from cro_amd.api.v1alpha1.types import DaemonSet
+    from cro_amd.nodeops.nodes import RESTARTED_AT_ANNOTATION, restart_daemonset
+
+    ds = DaemonSet()
+    ds.metadata.name = "ns/ds"
+    client.create(ds)  # desired == 0
+    assert restart_daemonset(client, "ns", "ds") is False
+
+    got = client.get(DaemonSet, "ns/ds")
+    got.status.desired_number_scheduled = 2
+    got.status.number_ready = 1  # rollout in flight
+    got.status.current_number_scheduled = 2
+    client.update_status(got)
+    assert restart_daemonset(client, "ns", "ds") is False
+
+    got = client.get(DaemonSet, "ns/ds")
+    got.status.number_ready = 2
+    client.update_status(got)
+    assert restart_daemonset(client, "ns", "ds") is True
+
+    got = client.get(DaemonSet, "ns/ds")
+    got.spec.template_annotations[RESTARTED_AT_ANNOTATION] = "not-a-time"
+    client.update(got)
+    with pytest.raises(ValueError, match="restartedAt"):
+        restart_daemonset(client, "ns", "ds")
+
+
+def test_trusted_binary_resolution(tmp_path):
+    """LocalNodeExec resolves binaries only from the trusted path list and
+    rejects path-qualified names (gpus.go:996-1038 trust model)."""
+    from cro_amd.nodeops.execs import ExecError, LocalNodeExec
+
+    ex = LocalNodeExec()
+    assert ex.resolve_binary("modprobe").startswith(("/usr/sbin", "/sbin", "/usr/bin"))
+    with pytest.raises(ExecError, match="bare"):
+        ex.resolve_binary("/tmp/evil/nvidia-smi")
+    with pytest.raises(ExecError, match="trusted"):
+        ex.resolve_binary("definitely-not-a-binary-xyz")
