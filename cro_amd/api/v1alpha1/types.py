@@ -158,7 +158,8 @@ class ComposableResource(K8sObject):
 
 
 class NodeCapacity(_Model):
-    """Allocatable capacity relevant to other_spec admission (nodes.go:78-117)."""
+    """Node Status.Capacity figures compared in other_spec admission
+    (nodes.go:84-117 uses Capacity, not Allocatable)."""
 
     milli_cpu: int = 0
     memory: int = 0
@@ -167,7 +168,7 @@ class NodeCapacity(_Model):
 
 
 class NodeStatus(_Model):
-    allocatable: NodeCapacity = Field(default_factory=NodeCapacity)
+    capacity: NodeCapacity = Field(default_factory=NodeCapacity)
     # amdgpu enumeration surfaced by the node agent (device-id -> render node)
     provider_id: str = ""
 

@@ -152,9 +152,9 @@ def build_local_stack(
 
     node = Node()
     node.metadata.name = node_name
-    node.status.allocatable.milli_cpu = 128000
-    node.status.allocatable.memory = 2 << 40
-    node.status.allocatable.allowed_pod_number = 256
+    node.status.capacity.milli_cpu = 128000
+    node.status.capacity.memory = 2 << 40
+    node.status.capacity.allowed_pod_number = 256
     mgr.client.create(node)
 
     stack = LocalStack(

@@ -4,7 +4,7 @@ Parity with internal/utils/nodes.go:
 
 * :func:`check_node_existed` / :func:`get_all_nodes` (nodes.go:119-144);
 * :func:`check_node_capacity_sufficient` — admission of ``other_spec``
-  CPU/memory/ephemeral/pods requirements against node allocatable
+  CPU/memory/ephemeral/pods requirements against node capacity
   (nodes.go:78-117), sized in deployments for 8 × MI355X / 288 GB HBM3E
   per node;
 * :class:`Debouncer` — the 10-second restart debounce the reference applies
@@ -43,7 +43,7 @@ def check_node_capacity_sufficient(
     client: Client, node_name: str, other_spec: NodeSpecRequirements
 ) -> bool:
     node = client.get(Node, node_name)
-    alloc = node.status.allocatable
+    alloc = node.status.capacity
     return (
         alloc.milli_cpu >= other_spec.milli_cpu
         and alloc.memory >= other_spec.memory

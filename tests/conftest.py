@@ -29,10 +29,10 @@ def make_node(client, name: str, milli_cpu=64000, memory=1 << 40, pods=128):
 
     n = Node()
     n.metadata.name = name
-    n.status.allocatable.milli_cpu = milli_cpu
-    n.status.allocatable.memory = memory
-    n.status.allocatable.ephemeral_storage = 1 << 40
-    n.status.allocatable.allowed_pod_number = pods
+    n.status.capacity.milli_cpu = milli_cpu
+    n.status.capacity.memory = memory
+    n.status.capacity.ephemeral_storage = 1 << 40
+    n.status.capacity.allowed_pod_number = pods
     return client.create(n)
 
 
