@@ -104,7 +104,9 @@ class ComposableResourceReconciler(Reconciler):
             self._set_error(resource, str(exc))
             raise
 
-    def _restart_plugin_daemonsets(self, resource: ComposableResource, fatal: bool) -> None:
+    def _restart_plugin_daemonsets(
+        self, resource: ComposableResource, fatal: bool
+    ) -> ComposableResource:
         """DEVICE_PLUGIN node refresh: roll the AMD device-plugin and
         metrics-exporter daemonsets (nvidia-device-plugin-daemonset +
         nvidia-dcgm parity, composableresource_controller.go:257-269).
@@ -130,9 +132,12 @@ class ComposableResourceReconciler(Reconciler):
                 log.warning("failed to restart %s/%s: %s", namespace, name, exc)
                 resource.status.error = str(exc)
                 try:
-                    self.client.update_status(resource)
+                    # keep the fresh resourceVersion so the caller's next
+                    # status write does not conflict pointlessly
+                    resource = self.client.update_status(resource)
                 except ConflictError:
                     pass
+        return resource
 
     def _persist_device_identity(
         self, name: str, device_id: str, cdi_device_id: str
@@ -185,6 +190,8 @@ class ComposableResourceReconciler(Reconciler):
         if taints.has_device_taint(self.client, resource):
             taints.delete_device_taint(self.client, resource)
         did = False
+        if resource.status.state == "Online":
+            self.metrics.devices_online.dec()  # GC skips the Detaching edge
         if resource.status.state != "Deleting":
             resource.status.state = "Deleting"
             resource.status.error = f"target node {resource.spec.target_node} not found"
@@ -261,7 +268,7 @@ class ComposableResourceReconciler(Reconciler):
                 self.node_ops.check_no_loads(node)
             except GPULoadsPresent as exc:
                 log.warning("gpu loads during attach on %s: %s", node, exc)
-            self._restart_plugin_daemonsets(resource, fatal=False)
+            resource = self._restart_plugin_daemonsets(resource, fatal=False)
         with self._phase("node_refresh"):
             self.node_ops.refresh_after_attach(node)
 
