@@ -120,16 +120,23 @@ def build_app(client: Client) -> FastAPI:
 
         async def gen():
             events = client.watch([cls.KIND])
-            for obj in client.list(cls):
-                yield _json.dumps({"type": "ADDED", "object": _dump(obj)}) + "\n"
-            loop = asyncio.get_running_loop()
-            while True:
-                try:
-                    ev = await loop.run_in_executor(None, events.get, True, 1.0)
-                except _queue.Empty:
-                    yield "\n"  # keepalive; also surfaces disconnects
-                    continue
-                yield _json.dumps({"type": ev.type, "object": _dump(ev.object)}) + "\n"
+            try:
+                for obj in client.list(cls):
+                    yield _json.dumps({"type": "ADDED", "object": _dump(obj)}) + "\n"
+                loop = asyncio.get_running_loop()
+                while True:
+                    try:
+                        ev = await loop.run_in_executor(None, events.get, True, 1.0)
+                    except _queue.Empty:
+                        yield "\n"  # keepalive; also surfaces disconnects
+                        continue
+                    yield _json.dumps({"type": ev.type, "object": _dump(ev.object)}) + "\n"
+            finally:
+                # disconnects must release the watcher or every later event
+                # fans out to dead queues forever
+                store = getattr(client, "store", None)
+                if store is not None and hasattr(store, "stop_watch"):
+                    store.stop_watch(events)
 
         return StreamingResponse(gen(), media_type="application/x-ndjson")
 

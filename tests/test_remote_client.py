@@ -137,3 +137,18 @@ def test_remote_operator_full_lifecycle(remote_operator):
     assert wait_for(lambda: remote.try_get(ComposabilityRequest, "r1") is None)
     assert wait_for(lambda: fabric.attached_to("node0") == [])
     assert server_mgr.client.list(ComposableResource) == []
+
+
+def test_watch_disconnect_releases_watcher(api_server):
+    """Server-side watchers must be unsubscribed when the stream client
+    disconnects — otherwise every event fans out to dead queues forever."""
+    url, server_mgr = api_server
+    before = len(server_mgr.store._watchers)
+    r1 = RemoteClient(url)
+    q = r1.watch(["ComposabilityRequest"])
+    # wait for the stream to register server-side
+    assert wait_for(lambda: len(server_mgr.store._watchers) > before, timeout=10)
+    r1.close()
+    assert wait_for(
+        lambda: len(server_mgr.store._watchers) == before, timeout=10
+    ), f"{len(server_mgr.store._watchers)} watchers still registered"
