@@ -169,10 +169,17 @@ class FTICMClient(FabricProvider):
             if spec_uuid:
                 break
         if failure:
-            # surfaced into status by the caller's error handling; the resize
-            # is still issued (cm/client.go:206-215 updates status and
-            # continues)
+            # persist the upstream failure reason and STILL issue the resize
+            # (cm/client.go:206-215: status update, then continue — the
+            # caller's Waiting requeue would otherwise drop the message)
             resource.status.error = failure
+            if self.client is not None:
+                try:
+                    fresh = self.client.get(ComposableResource, resource.metadata.name)
+                    fresh.status.error = failure
+                    self.client.update_status(fresh)
+                except Exception:  # best-effort, like the reference's log-and-go
+                    pass
         if not spec_uuid:
             return  # device already gone upstream — idempotent
 
