@@ -86,9 +86,19 @@ def build_app(client: Client) -> FastAPI:
         return {"status": "ok"}
 
     @app.get("/metrics")
-    def metrics():
+    def metrics(request: Request):
+        """Prometheus endpoint; CRO_METRICS_TOKEN enables bearer-token
+        authentication (the reference serves metrics behind an
+        authn/authz filter, cmd/main.go:109-127)."""
+        import os
+
         import prometheus_client
 
+        token = os.environ.get("CRO_METRICS_TOKEN", "")
+        if token:
+            auth = request.headers.get("authorization", "")
+            if auth != f"Bearer {token}":
+                raise HTTPException(401, "metrics require a valid bearer token")
         return Response(
             prometheus_client.generate_latest(),
             media_type=prometheus_client.CONTENT_TYPE_LATEST,

@@ -110,3 +110,12 @@ def test_metrics_endpoint(api_stack):
     resp = http.get("/metrics")
     assert resp.status_code == 200
     assert b"cro_reconcile_total" in resp.content
+
+
+def test_metrics_token_auth(api_stack, monkeypatch):
+    http, _ = api_stack
+    monkeypatch.setenv("CRO_METRICS_TOKEN", "sekrit")
+    assert http.get("/metrics").status_code == 401
+    assert http.get("/metrics", headers={"Authorization": "Bearer wrong"}).status_code == 401
+    ok = http.get("/metrics", headers={"Authorization": "Bearer sekrit"})
+    assert ok.status_code == 200 and b"cro_reconcile_total" in ok.content
