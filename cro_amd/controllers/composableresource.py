@@ -237,9 +237,10 @@ class ComposableResourceReconciler(Reconciler):
 
         node = resource.spec.target_node
         mode = self.adapter.device_resource_type
+        ops = self.node_ops.for_type(resource.spec.type)
 
         with self._phase("driver_gate"):
-            self.node_ops.ensure_driver(node)
+            ops.ensure_driver(node)
 
         if resource.status.device_id == "":
             t0 = time.monotonic()
@@ -265,17 +266,17 @@ class ComposableResourceReconciler(Reconciler):
             # load check is advisory on attach (reference logs and continues,
             # composableresource_controller.go:253-256)
             try:
-                self.node_ops.check_no_loads(node)
+                ops.check_no_loads(node)
             except GPULoadsPresent as exc:
                 log.warning("gpu loads during attach on %s: %s", node, exc)
             resource = self._restart_plugin_daemonsets(resource, fatal=False)
         with self._phase("node_refresh"):
-            self.node_ops.refresh_after_attach(node)
+            ops.refresh_after_attach(node)
 
         if mode == "DRA":
-            visible = self.node_ops.is_visible_dra(node, resource.status.device_id)
+            visible = ops.is_visible_dra(node, resource.status.device_id)
         else:
-            visible = self.node_ops.is_visible(node, resource.status.device_id)
+            visible = ops.is_visible(node, resource.status.device_id)
         if not visible:
             return Result(requeue_after=self.config.attach_visible_wait)
 
@@ -285,9 +286,9 @@ class ComposableResourceReconciler(Reconciler):
         # (FM detach keys on it, fm/client.go:231-242) — the container-CDI
         # spec name is amd.com/gpu=<device_id>, never stored over it.
         with self._phase("cdi_write"):
-            self.node_ops.write_cdi(node, resource.status.device_id)
+            ops.write_cdi(node, resource.status.device_id)
         with self._phase("health_probe"):
-            probe = self.node_ops.health_probe(node, resource.status.device_id)
+            probe = ops.health_probe(node, resource.status.device_id)
         if probe is not None and not probe.get("ok", True):
             raise FabricError(f"gfx950 health probe failed: {probe}")
 
@@ -333,19 +334,20 @@ class ComposableResourceReconciler(Reconciler):
     def _handle_detaching(self, resource: ComposableResource) -> Result:
         node = resource.spec.target_node
         mode = self.adapter.device_resource_type
+        ops = self.node_ops.for_type(resource.spec.type)
 
         if resource.status.device_id != "":
             if not resource.spec.force_detach:
                 if mode == "DEVICE_PLUGIN":
-                    self.node_ops.check_no_loads(node)  # whole node
+                    ops.check_no_loads(node)  # whole node
                 else:
-                    self.node_ops.check_no_loads(node, resource.status.device_id)
+                    ops.check_no_loads(node, resource.status.device_id)
 
             if mode == "DRA":
                 taints.create_device_taint(self.client, resource)
 
             try:
-                self.node_ops.drain(node, resource.status.device_id)
+                ops.drain(node, resource.status.device_id)
             except amdgpu.DrainInProgress:
                 # last-device drains run asynchronously (module unload can
                 # block); poll completion at detach-wait granularity
@@ -364,16 +366,16 @@ class ComposableResourceReconciler(Reconciler):
             if mode == "DEVICE_PLUGIN":
                 # fatal on detach (composableresource_controller.go:379-385)
                 self._restart_plugin_daemonsets(resource, fatal=True)
-            self.node_ops.refresh_after_detach(node)
+            ops.refresh_after_detach(node)
 
             if mode == "DRA":
-                visible = self.node_ops.is_visible_dra(node, resource.status.device_id)
+                visible = ops.is_visible_dra(node, resource.status.device_id)
             else:
-                visible = self.node_ops.is_visible(node, resource.status.device_id)
+                visible = ops.is_visible(node, resource.status.device_id)
             if visible:
                 return Result(requeue_after=self.config.detach_invisible_wait)
 
-            self.node_ops.remove_cdi(node, resource.status.device_id)
+            ops.remove_cdi(node, resource.status.device_id)
 
             if mode == "DRA":
                 taints.delete_device_taint(self.client, resource)

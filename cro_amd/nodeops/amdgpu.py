@@ -56,6 +56,11 @@ class DrainInProgress(Exception):
 class NodeOps:
     """Surface the ComposableResource controller programs against."""
 
+    def for_type(self, resource_type: str) -> "NodeOps":
+        """Per-resource-type dispatch hook (CompositeNodeOps overrides);
+        single-type implementations serve every type themselves."""
+        return self
+
     def ensure_driver(self, node: str) -> None:
         raise NotImplementedError  # pragma: no cover
 

@@ -34,13 +34,26 @@ def cdi_device_id(device_id: str) -> str:
 
 
 class CDISpecWriter:
-    def __init__(self, execer: NodeExec, cdi_dir: str = DEFAULT_CDI_DIR):
+    def __init__(
+        self,
+        execer: NodeExec,
+        cdi_dir: str = DEFAULT_CDI_DIR,
+        kind: str = CDI_KIND,
+        root_device_nodes=None,
+    ):
+        """``kind`` names the CDI device class (one spec file per kind);
+        ``root_device_nodes`` are class-wide device nodes injected for every
+        device (GPUs: /dev/kfd; CXL.mem: none)."""
         self.execer = execer
         self.cdi_dir = cdi_dir.rstrip("/")
+        self.kind = kind
+        self.root_device_nodes = (
+            root_device_nodes if root_device_nodes is not None else ["/dev/kfd"]
+        )
         self._lock = threading.Lock()
 
     def _spec_path(self, node: str) -> str:
-        return f"{self.cdi_dir}/{CDI_KIND.replace('/', '-')}-cro.json"
+        return f"{self.cdi_dir}/{self.kind.replace('/', '-')}-cro.json"
 
     def _load(self, node: str) -> dict:
         try:
@@ -48,8 +61,10 @@ class CDISpecWriter:
         except (FileNotFoundError, json.JSONDecodeError):
             return {
                 "cdiVersion": CDI_VERSION,
-                "kind": CDI_KIND,
-                "containerEdits": {"deviceNodes": [{"path": "/dev/kfd"}]},
+                "kind": self.kind,
+                "containerEdits": {
+                    "deviceNodes": [{"path": p} for p in self.root_device_nodes]
+                },
                 "devices": [],
             }
 
@@ -81,7 +96,26 @@ class CDISpecWriter:
             spec["devices"] = [d for d in spec["devices"] if d["name"] != gpu.device_id]
             spec["devices"].append(entry)
             self._store(node, spec)
-            return cdi_device_id(gpu.device_id)
+            return f"{self.kind}={gpu.device_id}"
+
+    def add_raw_device(
+        self, node: str, device_id: str, device_nodes, annotations
+    ) -> str:
+        """Add a device entry from explicit nodes/annotations (non-GPU
+        device classes); returns its CDI device id."""
+        with self._lock:
+            spec = self._load(node)
+            entry = {
+                "name": device_id,
+                "containerEdits": {
+                    "deviceNodes": [{"path": p} for p in device_nodes],
+                },
+                "annotations": dict(annotations),
+            }
+            spec["devices"] = [d for d in spec["devices"] if d["name"] != device_id]
+            spec["devices"].append(entry)
+            self._store(node, spec)
+            return f"{self.kind}={device_id}"
 
     def remove_device(self, node: str, device_id: str) -> None:
         with self._lock:
