@@ -8,6 +8,7 @@
 // JSON on stdout.
 //
 //   croagent list   [--sysroot /]            GPU inventory (KFD topology)
+//   croagent cxl    [--sysroot /]            CXL.mem inventory (/sys/bus/cxl)
 //   croagent pids   [--gpu-id N]             KFD compute processes
 //   croagent probe  [--device N]             gfx950 MFMA/HBM health probe
 //   croagent drain  --bdf 0000:5a:00.0       sysfs PCI remove
@@ -179,6 +180,41 @@ int cmd_list() {
   return 0;
 }
 
+int cmd_cxl() {
+  const std::string base = "/sys/bus/cxl/devices";
+  printf("{\"memdevs\":[");
+  bool first = true;
+  for (const auto& entry : list_dir(base)) {
+    if (entry.rfind("mem", 0) != 0 || entry.size() <= 3 || !isdigit(entry[3])) continue;
+    std::string text;
+    unsigned long long serial = 0, size = 0;
+    long long numa = 0;
+    if (read_file(base + "/" + entry + "/serial", &text))
+      serial = strtoull(text.c_str(), nullptr, 16);
+    if (read_file(base + "/" + entry + "/ram/size", &text))
+      size = strtoull(text.c_str(), nullptr, 16);
+    if (read_file(base + "/" + entry + "/numa_node", &text)) numa = atoll(text.c_str());
+    std::string bdf;
+    if (read_file(base + "/" + entry + "/device/uevent", &text)) {
+      std::istringstream in(text);
+      std::string line;
+      while (std::getline(in, line)) {
+        const std::string kPrefix = "PCI_SLOT_NAME=";
+        if (line.rfind(kPrefix, 0) == 0) bdf = line.substr(kPrefix.size());
+      }
+    }
+    char id[64];
+    if (serial) snprintf(id, sizeof(id), "CXL-%016llx", serial);
+    else snprintf(id, sizeof(id), "CXL-pci-%s", bdf.c_str());
+    printf("%s{\"device_id\":\"%s\",\"memdev\":\"%s\",\"size_bytes\":%llu,"
+           "\"numa_node\":%lld,\"pci_bdf\":\"%s\"}",
+           first ? "" : ",", id, entry.c_str(), size, numa, bdf.c_str());
+    first = false;
+  }
+  printf("]}\n");
+  return 0;
+}
+
 int cmd_pids(long long gpu_id) {
   const std::string base = "/sys/class/kfd/kfd/proc";
   printf("{\"pids\":[");
@@ -237,7 +273,7 @@ int cmd_probe(int device) {
 
 int main(int argc, char** argv) {
   if (argc < 2) {
-    fprintf(stderr, "usage: croagent <list|pids|probe|drain|rescan> [options]\n");
+    fprintf(stderr, "usage: croagent <list|cxl|pids|probe|drain|rescan> [options]\n");
     return 2;
   }
   std::string cmd = argv[1];
@@ -252,6 +288,7 @@ int main(int argc, char** argv) {
     else if (arg == "--device" && i + 1 < argc) device = atoi(argv[++i]);
   }
   if (cmd == "list") return cmd_list();
+  if (cmd == "cxl") return cmd_cxl();
   if (cmd == "pids") return cmd_pids(gpu_id);
   if (cmd == "probe") return cmd_probe(device);
   if (cmd == "drain") return cmd_drain(bdf);

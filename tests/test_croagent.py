@@ -113,3 +113,28 @@ def test_probe_on_real_gpu(agent):
     data = json.loads(proc.stdout)
     assert data["ok"] and data["mfma_f32_exact"]
     assert "gfx950" in data["gcn_arch"]
+
+
+def test_cxl_inventory(agent, tmp_path):
+    base = tmp_path / "sys" / "bus" / "cxl" / "devices" / "mem0"
+    base.mkdir(parents=True)
+    (base / "serial").write_text("0xc0ffee00\n")
+    ram = base / "ram"
+    ram.mkdir()
+    (ram / "size").write_text("0x4000000000\n")
+    (base / "numa_node").write_text("2\n")
+    dev = base / "device"
+    dev.mkdir()
+    (dev / "uevent").write_text("DRIVER=cxl_pci\nPCI_SLOT_NAME=0000:60:00.0\n")
+    proc = run_agent(agent, "cxl", "--sysroot", str(tmp_path))
+    assert proc.returncode == 0, proc.stderr
+    data = json.loads(proc.stdout)
+    assert data["memdevs"] == [
+        {
+            "device_id": f"CXL-{0xC0FFEE00:016x}",
+            "memdev": "mem0",
+            "size_bytes": 0x4000000000,
+            "numa_node": 2,
+            "pci_bdf": "0000:60:00.0",
+        }
+    ]
