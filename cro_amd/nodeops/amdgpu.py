@@ -127,6 +127,9 @@ class AmdNodeOps(NodeOps):
         self.enum_cache_ttl = 0.025
         self._enum_cache: Dict[str, tuple] = {}  # node -> (monotonic, gpus)
         self._enum_lock = threading.Lock()
+        # per-device static metadata (VRAM/xGMI/card map) survives the TTL
+        # cache — keyed by device identity, see kfd.enumerate_gpus
+        self._static_meta: dict = {}
 
     # -- driver ------------------------------------------------------------
 
@@ -151,7 +154,7 @@ class AmdNodeOps(NodeOps):
                 gpus = None
         if gpus is None:
             try:
-                gpus = enumerate_gpus(self.execer, node)
+                gpus = enumerate_gpus(self.execer, node, self._static_meta)
             except ExecError:
                 gpus = enumerate_gpus_amdsmi(self.execer, node)
             with self._enum_lock:

@@ -111,13 +111,28 @@ def _drm_card_by_bdf(execer: NodeExec, node: str) -> Dict[str, int]:
     return mapping
 
 
-def enumerate_gpus(execer: NodeExec, node: str) -> List[GPUDevice]:
-    """Enumerate GPUs from KFD topology; raises ExecError if KFD is absent."""
+def enumerate_gpus(
+    execer: NodeExec, node: str, static_cache: Optional[dict] = None
+) -> List[GPUDevice]:
+    """Enumerate GPUs from KFD topology; raises ExecError if KFD is absent.
+
+    ``static_cache`` (owned by the caller, e.g. AmdNodeOps) memoizes the
+    per-device data that cannot change while a device keeps its identity —
+    VRAM bank sizes, xGMI peers and the DRM card index — so steady-state
+    re-enumeration reads only each node's ``properties``/``gpu_id``.  xGMI
+    topology is chassis-fixed; a device re-composed under a new unique_id
+    misses the cache and re-reads everything.
+    """
     try:
         entries = execer.list_dir(node, KFD_NODES)
     except (FileNotFoundError, PermissionError, OSError):
         raise ExecError("KFD topology not present (amdgpu driver not loaded?)")
-    card_map = _drm_card_by_bdf(execer, node)
+    if static_cache is None:
+        static_cache = {}
+    card_map = static_cache.get("_card_map")
+    if card_map is None:
+        card_map = _drm_card_by_bdf(execer, node)
+        static_cache["_card_map"] = card_map
 
     gpus: List[GPUDevice] = []
     for entry in entries:
@@ -148,6 +163,12 @@ def enumerate_gpus(execer: NodeExec, node: str) -> List[GPUDevice]:
             gfx_target=str(props.get("gfx_target_version", "")),
             card_index=card_map.get(bdf.lower()),
         )
+        cache_key = (entry, unique_id, bdf)
+        cached = static_cache.get(cache_key)
+        if cached is not None:
+            dev.vram_bytes, dev.xgmi_peers = cached[0], list(cached[1])
+            gpus.append(dev)
+            continue
         # VRAM from mem_banks (heap_type 1/2 = FB public/private)
         try:
             for bank in execer.list_dir(node, f"{base}/mem_banks"):
@@ -168,6 +189,7 @@ def enumerate_gpus(execer: NodeExec, node: str) -> List[GPUDevice]:
                     dev.xgmi_peers.append(lprops.get("node_to", -1))
         except (FileNotFoundError, PermissionError, OSError):
             pass
+        static_cache[cache_key] = (dev.vram_bytes, list(dev.xgmi_peers))
         gpus.append(dev)
     return gpus
 
