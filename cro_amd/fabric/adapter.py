@@ -37,6 +37,11 @@ def new_adapter(client=None, provider: FabricProvider = None) -> Adapter:
     if provider is not None:  # injected (tests, bench, embedded mock)
         return Adapter(device_resource_type, provider)
 
+    # CRO_FABRIC_TLS_VERIFY=false disables certificate verification toward
+    # the fabric manager (the reference tests' InsecureSkipVerify seam,
+    # composableresource_controller_test.go:999-1005; production keeps it on)
+    verify = os.environ.get("CRO_FABRIC_TLS_VERIFY", "true").lower() != "false"
+
     provider_type = os.environ.get("CDI_PROVIDER_TYPE", "")
     if provider_type == "MOCK":
         from .mock import MockFabric
@@ -60,11 +65,11 @@ def new_adapter(client=None, provider: FabricProvider = None) -> Adapter:
         if api_type == "CM":
             from .fti.cm import FTICMClient
 
-            provider = FTICMClient(client)
+            provider = FTICMClient(client, verify=verify)
         elif api_type == "FM":
             from .fti.fm import FTIFMClient
 
-            provider = FTIFMClient(client)
+            provider = FTIFMClient(client, verify=verify)
         else:
             raise ValueError(
                 f"the env variable FTI_CDI_API_TYPE has an invalid value: '{api_type}'"

@@ -79,3 +79,14 @@ def test_nec_selected(monkeypatch):
     monkeypatch.setenv("LAYOUT_APPLY_PORT", "8000")
     monkeypatch.setenv("CONFIGURATION_MANAGER_PORT", "8001")
     assert new_adapter().provider.name == "nec"
+
+
+def test_fabric_tls_verify_knob(monkeypatch):
+    monkeypatch.setenv("DEVICE_RESOURCE_TYPE", "DRA")
+    monkeypatch.setenv("CDI_PROVIDER_TYPE", "FTI_CDI")
+    monkeypatch.setenv("FTI_CDI_CLUSTER_ID", "cluster-1")
+    monkeypatch.setenv("FTI_CDI_ENDPOINT", "fabric.example")
+    monkeypatch.setenv("FTI_CDI_API_TYPE", "FM")
+    assert new_adapter().provider.verify is True
+    monkeypatch.setenv("CRO_FABRIC_TLS_VERIFY", "false")
+    assert new_adapter().provider.verify is False
