@@ -40,6 +40,18 @@ _COLUMNS = {
         ("ERROR", lambda o: (o["status"].get("error", "") or "")[:40]),
     ),
 }
+_COLUMNS["nodes"] = (
+    ("NAME", lambda o: o["metadata"]["name"]),
+    ("PROVIDER-ID", lambda o: o["status"].get("provider_id", "")),
+    ("CPU(m)", lambda o: str(o["status"]["capacity"]["milli_cpu"])),
+    ("PODS", lambda o: str(o["status"]["capacity"]["allowed_pod_number"])),
+)
+_COLUMNS["resourceslices"] = (
+    ("NAME", lambda o: o["metadata"]["name"]),
+    ("NODE", lambda o: o["spec"]["node_name"]),
+    ("DRIVER", lambda o: o["spec"]["driver"]),
+    ("DEVICES", lambda o: str(len(o["spec"]["devices"]))),
+)
 _DEFAULT_COLUMNS = (
     ("NAME", lambda o: o["metadata"]["name"]),
     ("KIND", lambda o: o.get("kind", "")),

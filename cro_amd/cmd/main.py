@@ -59,12 +59,32 @@ def main(argv=None) -> int:
     p.add_argument("--max-concurrent-reconciles", type=int, default=8)
     p.add_argument("--syncer-period", type=float, default=60.0)
     p.add_argument("--zap-log-level", default="info")
+    p.add_argument("--log-format", choices=["text", "json"], default="text")
     args = p.parse_args(argv)
 
-    logging.basicConfig(
-        level=getattr(logging, args.zap_log_level.upper(), logging.INFO),
-        format="%(asctime)s %(levelname)s %(name)s %(message)s",
-    )
+    if args.log_format == "json":
+        import json as _json
+
+        class JsonFormatter(logging.Formatter):
+            def format(self, record):
+                return _json.dumps({
+                    "ts": self.formatTime(record),
+                    "level": record.levelname,
+                    "logger": record.name,
+                    "msg": record.getMessage(),
+                })
+
+        handler = logging.StreamHandler()
+        handler.setFormatter(JsonFormatter())
+        logging.basicConfig(
+            level=getattr(logging, args.zap_log_level.upper(), logging.INFO),
+            handlers=[handler],
+        )
+    else:
+        logging.basicConfig(
+            level=getattr(logging, args.zap_log_level.upper(), logging.INFO),
+            format="%(asctime)s %(levelname)s %(name)s %(message)s",
+        )
     log = logging.getLogger("cro_amd.main")
 
     lock_file = None
