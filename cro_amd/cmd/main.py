@@ -124,12 +124,24 @@ def main(argv=None) -> int:
             probe_fn = probe_fn_for_nodeops
     except Exception:
         log.warning("gfx950 probe library unavailable; health probe disabled")
-    node_ops = AmdNodeOps(
-        LocalNodeExec(),
+    execer = LocalNodeExec()
+    gpu_ops = AmdNodeOps(
+        execer,
         client=mgr.client,
         cdi_dir=args.cdi_dir,
         destructive=args.destructive,
         probe_fn=probe_fn,
+    )
+    from ..nodeops.composite import CompositeNodeOps
+    from ..nodeops.cxl import CxlNodeOps
+
+    node_ops = CompositeNodeOps(
+        {
+            "gpu": gpu_ops,
+            "cxlmemory": CxlNodeOps(
+                execer, cdi_dir=args.cdi_dir, destructive=args.destructive
+            ),
+        }
     )
     mgr.resource_reconciler.node_ops = node_ops
     if hasattr(mgr, "syncer"):
