@@ -414,3 +414,46 @@ def test_trusted_binary_resolution(tmp_path):
         ex.resolve_binary("/tmp/evil/nvidia-smi")
     with pytest.raises(ExecError, match="trusted"):
         ex.resolve_binary("definitely-not-a-binary-xyz")
+
+
+def test_amdsmi_fallback_via_node_ops():
+    """KFD missing entirely → AmdNodeOps falls back to amd-smi enumeration."""
+    ex = MockNodeExec()
+    ex.set_file(NODE, "/sys/module/amdgpu/version", "6.x")
+    ex.set_command(
+        ("amd-smi", "list", "--json"),
+        (0, '[{"gpu": 0, "bdf": "0000:0a:00.0", '
+            '"uuid": "deadbeef-0000-1000-8000-000000000000", '
+            '"kfd_id": 111, "node_id": 2, "partition_id": 0}]', ""),
+    )
+    ops = AmdNodeOps(ex)
+    gpus = ops.enumerate(NODE)
+    assert len(gpus) == 1 and gpus[0].pci_bdf == "0000:0a:00.0"
+
+
+def test_enum_ttl_cache_and_invalidation():
+    ex = MockNodeExec()
+    ids = kfd_fixture(ex, 1)
+    ops = AmdNodeOps(ex, destructive=False)
+    ops.enum_cache_ttl = 10.0  # effectively permanent for this test
+    assert ops.is_visible(NODE, ids[0])
+    reads_before = len(ex.calls)
+    ops.is_visible(NODE, ids[0])  # served from cache
+    assert len(ex.calls) == reads_before
+    ops.drain(NODE, ids[0])  # lifecycle mutation invalidates
+    assert not ops.is_visible(NODE, ids[0])
+    assert len(ex.calls) > reads_before  # re-read happened
+
+
+def test_static_meta_survives_invalidation():
+    ex = MockNodeExec()
+    ids = kfd_fixture(ex, 1)
+    ops = AmdNodeOps(ex, destructive=False)
+    ops.enumerate(NODE)
+    mem_reads = [c for c in ex.calls if c[0] == "read" and "mem_banks" in c[2]]
+    ops._invalidate_enum(NODE)
+    ops.enumerate(NODE)
+    mem_reads_after = [c for c in ex.calls if c[0] == "read" and "mem_banks" in c[2]]
+    # vram banks were NOT re-read (static cache), yet values are present
+    assert len(mem_reads_after) == len(mem_reads)
+    assert ops.enumerate(NODE)[0].vram_bytes == 309237645312
