@@ -147,6 +147,42 @@ def main(argv=None) -> int:
     if hasattr(mgr, "syncer"):
         mgr.syncer.node_ops = node_ops
 
+    # MOCK fabric + a local node: bind the pool to the real inventory so
+    # attaches hand out devices that actually exist (the bench-harness
+    # binding, made available to the production entrypoint for demos and
+    # single-node operation without a physical fabric)
+    if os.environ.get("CDI_PROVIDER_TYPE", "") == "MOCK" and args.node and not args.destructive:
+        try:
+            gpus = gpu_ops.enumerate(args.node)
+        except Exception:
+            gpus = []
+        if gpus:
+            from ..fabric.mock import MockFabric
+
+            fabric = MockFabric(
+                bind_inventory=[
+                    {
+                        "device_id": g.device_id,
+                        "cdi_device_id": f"amd.com/gpu={g.device_id}",
+                        "model": "mi355x",
+                    }
+                    for g in gpus
+                ]
+            )
+            gpu_ops._sim_detached = {g.device_id for g in gpus}
+            gpu_ops._invalidate_enum(args.node)
+
+            orig_add = fabric.add_resource
+
+            def _bound_add(resource):
+                did, cdi = orig_add(resource)
+                gpu_ops.simulate_compose(resource.spec.target_node, did)
+                return did, cdi
+
+            fabric.add_resource = _bound_add
+            adapter.provider = fabric
+            log.info("MOCK fabric bound to %d local device(s)", len(gpus))
+
     if not args.serve_only:
         mgr.start()
         log.info("manager started (%d reconcile workers per controller)",
