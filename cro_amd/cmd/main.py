@@ -147,6 +147,31 @@ def main(argv=None) -> int:
     if hasattr(mgr, "syncer"):
         mgr.syncer.node_ops = node_ops
 
+    # standalone mode owns the store, so the local node registers itself
+    # (cluster mode gets Nodes from the apiserver)
+    if remote is None and args.node:
+        from ..api.v1alpha1.types import Node
+        from ..runtime.errors import AlreadyExistsError
+
+        node_obj = Node()
+        node_obj.metadata.name = args.node
+        node_obj.status.capacity.milli_cpu = (os.cpu_count() or 1) * 1000
+        try:
+            with open("/proc/meminfo") as f:
+                for line in f:
+                    if line.startswith("MemTotal:"):
+                        node_obj.status.capacity.memory = int(line.split()[1]) * 1024
+                        break
+        except OSError:
+            pass
+        node_obj.status.capacity.allowed_pod_number = 110
+        node_obj.status.capacity.ephemeral_storage = 1 << 40
+        try:
+            mgr.client.create(node_obj)
+            log.info("registered local node %s", args.node)
+        except AlreadyExistsError:
+            pass
+
     # MOCK fabric + a local node: bind the pool to the real inventory so
     # attaches hand out devices that actually exist (the bench-harness
     # binding, made available to the production entrypoint for demos and
