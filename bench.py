@@ -41,6 +41,10 @@ def main() -> int:
     p.add_argument("--size", type=int, default=1, help="devices per request")
     p.add_argument("--mode", default="DRA", choices=["DRA", "DEVICE_PLUGIN"])
     p.add_argument("--no-probe", action="store_true")
+    p.add_argument("--fabric-latency", type=float, default=0.0,
+                   help="simulated fabric compose/decompose seconds")
+    p.add_argument("--fabric-async", action="store_true",
+                   help="CM-style asynchronous fabric (resize+poll)")
     p.add_argument(
         "--force-detach",
         action="store_true",
@@ -75,12 +79,22 @@ def main() -> int:
     )
 
     node_name = f"bench-node-r{rank}"
+    fabric_config = None
+    if args.fabric_latency > 0 or args.fabric_async:
+        from cro_amd.fabric.mock import MockFabricConfig
+
+        fabric_config = MockFabricConfig(
+            attach_latency=args.fabric_latency,
+            detach_latency=args.fabric_latency,
+            asynchronous=args.fabric_async,
+        )
     stack = build_local_stack(
         node_name=node_name,
         mode=args.mode,
         use_gpu=use_gpu,
         gpu_index=local_rank if use_gpu else None,
         enable_probe=use_gpu and not args.no_probe,
+        fabric_config=fabric_config,
         cdi_dir=os.path.join(
             os.environ.get("TMPDIR", "/tmp"), f"cro-cdi-bench-r{rank}"
         ),
@@ -176,7 +190,10 @@ def main() -> int:
                 "global_batch": world * args.size,
                 "seq_len": 0,
                 "parallelism": f"one operator per GPU x{world}, 8 reconcile workers each",
-                "fabric": "mock (in-process; no physical CXL fabric on bench node)",
+                "fabric": "mock (in-process; no physical CXL fabric on bench node)"
+                + (f", simulated latency {args.fabric_latency}s"
+                   + (" async" if args.fabric_async else "")
+                   if args.fabric_latency else ""),
                 "node_path": "real KFD/CDI/HIP-probe" if use_gpu else "mock",
                 "device_resource_type": args.mode,
                 "probe": bool(use_gpu and not args.no_probe),

@@ -49,8 +49,12 @@ MANAGED_BY_LABEL = "app.kubernetes.io/managed-by"
 class ReconcileConfig:
     # wait for device visibility after fabric attach (reference: 30 s)
     attach_visible_wait: float = 0.05
-    # wait while fabric reports attach/detach in progress (reference: 30 s)
-    fabric_wait: float = 0.25
+    # waits while the fabric reports attach/detach in progress
+    # (reference: fixed 30 s): exponential from base to max so attach
+    # tracks the fabric's actual compose time + a bounded overshoot
+    # instead of quantizing to a fixed poll step
+    fabric_wait_base: float = 0.05
+    fabric_wait_max: float = 1.0
     # Online health-check period (reference: 30 s — not latency-critical)
     online_health_period: float = 30.0
     # wait for device invisibility during detach (reference: 3 s)
@@ -74,6 +78,8 @@ class ComposableResourceReconciler(Reconciler):
         # never corrupts it — the histogram only sees fully observed attaches)
         self._attach_started: Dict[str, float] = {}
         self._detach_started: Dict[str, float] = {}
+        # uid → consecutive fabric-waiting polls (for the exponential wait)
+        self._fabric_polls: Dict[str, int] = {}
 
     # -- plumbing ----------------------------------------------------------
 
@@ -138,6 +144,12 @@ class ComposableResourceReconciler(Reconciler):
                 except ConflictError:
                     pass
         return resource
+
+    def _fabric_wait(self, resource: ComposableResource) -> float:
+        n = self._fabric_polls.get(resource.metadata.uid, 0)
+        self._fabric_polls[resource.metadata.uid] = n + 1
+        return min(self.config.fabric_wait_base * (2 ** min(n, 16)),
+                   self.config.fabric_wait_max)
 
     def _persist_device_identity(
         self, name: str, device_id: str, cdi_device_id: str
@@ -247,7 +259,7 @@ class ComposableResourceReconciler(Reconciler):
             try:
                 device_id, cdi_device_id = self.adapter.provider.add_resource(resource)
             except WaitingDeviceAttaching:
-                return Result(requeue_after=self.config.fabric_wait)
+                return Result(requeue_after=self._fabric_wait(resource))
             finally:
                 self.metrics.fabric_request_seconds.labels(
                     self.adapter.provider.name, "add"
@@ -261,6 +273,7 @@ class ComposableResourceReconciler(Reconciler):
             resource = self._persist_device_identity(
                 resource.metadata.name, device_id, cdi_device_id
             )
+            self._fabric_polls.pop(resource.metadata.uid, None)
 
         if mode == "DEVICE_PLUGIN":
             # load check is advisory on attach (reference logs and continues,
@@ -357,7 +370,7 @@ class ComposableResourceReconciler(Reconciler):
             try:
                 self.adapter.provider.remove_resource(resource)
             except WaitingDeviceDetaching:
-                return Result(requeue_after=self.config.fabric_wait)
+                return Result(requeue_after=self._fabric_wait(resource))
             finally:
                 self.metrics.fabric_request_seconds.labels(
                     self.adapter.provider.name, "remove"
@@ -380,6 +393,7 @@ class ComposableResourceReconciler(Reconciler):
             if mode == "DRA":
                 taints.delete_device_taint(self.client, resource)
 
+            self._fabric_polls.pop(resource.metadata.uid, None)
             resource.status.error = ""
             resource.status.device_id = ""
             resource.status.cdi_device_id = ""

@@ -155,3 +155,14 @@ def test_node_deletion_garbage_collects_everything(stack):
     assert stack.mgr.wait_for(
         lambda: stack.mgr.client.list(ComposableResource) == [], timeout=15
     )
+
+
+def test_async_fabric_exponential_polling(stack):
+    """The fabric wait grows exponentially (50 ms → 1 s cap), so attach
+    tracks the fabric's compose time with bounded overshoot instead of
+    quantizing to a fixed step."""
+    stack.fabric.config = MockFabricConfig(asynchronous=True, attach_latency=0.3)
+    timing = attach_detach_cycle(stack, "expo-r", size=1, timeout=30)
+    # 0.3 s compose: exponential checks land ≈0.35-0.45 s; a fixed 250 ms
+    # step would land at ≈0.5 s and the reference at up to 30 s
+    assert 300 <= timing["attach_ms"] < 490, timing
