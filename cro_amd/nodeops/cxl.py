@@ -23,7 +23,7 @@ ROADMAP.md).
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 from .execs import ExecError, NodeExec
 
