@@ -140,6 +140,8 @@ def build_app(client: Client) -> FastAPI:
                     except _queue.Empty:
                         yield "\n"  # keepalive; also surfaces disconnects
                         continue
+                    except RuntimeError:
+                        return  # event loop / executor shutting down
                     yield _json.dumps({"type": ev.type, "object": _dump(ev.object)}) + "\n"
             finally:
                 # disconnects must release the watcher or every later event

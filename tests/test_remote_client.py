@@ -152,3 +152,84 @@ def test_watch_disconnect_releases_watcher(api_server):
     assert wait_for(
         lambda: len(server_mgr.store._watchers) == before, timeout=10
     ), f"{len(server_mgr.store._watchers)} watchers still registered"
+
+
+def test_api_server_outage_mid_lifecycle_recovers():
+    """The API server dies mid-attach and comes back (same store): the
+    remote operator's watch streams reconnect (re-list replay) and write
+    retries back off — the lifecycle converges."""
+    import uvicorn
+
+    from cro_amd.fabric.mock import MockFabricConfig
+
+    server_mgr = build_manager(Adapter("DRA", MockFabric()), None)
+    port = free_port()
+    app = build_app(server_mgr.client)
+
+    def start_server():
+        server = uvicorn.Server(
+            uvicorn.Config(app, host="127.0.0.1", port=port, log_level="error")
+        )
+        thread = threading.Thread(target=server.run, daemon=True)
+        thread.start()
+        import httpx
+
+        deadline = time.monotonic() + 15
+        while time.monotonic() < deadline:
+            try:
+                if httpx.get(f"http://127.0.0.1:{port}/healthz", timeout=1).status_code == 200:
+                    return server, thread
+            except Exception:
+                time.sleep(0.05)
+        raise AssertionError("server did not come up")
+
+    server, thread = start_server()
+
+    remote = RemoteClient(f"http://127.0.0.1:{port}")
+    # slow async fabric so the outage lands mid-attach
+    fabric = MockFabric(
+        models={"mi355x": 8},
+        config=MockFabricConfig(asynchronous=True, attach_latency=1.0),
+    )
+    mgr = build_manager(Adapter("DRA", fabric), None, client=remote, enable_webhook=False)
+    ops = MockNodeOps(client=mgr.client)
+    mgr.resource_reconciler.node_ops = ops
+
+    orig_add = fabric.add_resource
+
+    def add_resource(resource):
+        did, cdi = orig_add(resource)
+        ops.fabric_composed(resource.spec.target_node, did)
+        return did, cdi
+
+    fabric.add_resource = add_resource
+
+    node = Node()
+    node.metadata.name = "node0"
+    remote.create(node)
+    ops.set_driver("node0", True)
+    mgr.start()
+    try:
+        remote.create(make_request("r1", size=1, target_node="node0"))
+        time.sleep(0.3)  # mid-attach (fabric still composing)
+
+        server.should_exit = True
+        thread.join(timeout=10)
+        time.sleep(1.0)  # outage window: workers hit connection errors
+
+        server, thread = start_server()  # same app → same store
+
+        assert wait_for(
+            lambda: (req := remote.try_get(ComposabilityRequest, "r1")) is not None
+            and req.status.state == "Running",
+            timeout=30,
+        ), (remote.try_get(ComposabilityRequest, "r1") or object()).__dict__
+
+        remote.delete(ComposabilityRequest, "r1")
+        assert wait_for(lambda: remote.try_get(ComposabilityRequest, "r1") is None, timeout=20)
+        assert fabric.attached_to("node0") == []
+    finally:
+        mgr.stop()
+        remote.close()
+        server.should_exit = True
+        thread.join(timeout=5)
