@@ -10,7 +10,7 @@
 //   croagent list   [--sysroot /]            GPU inventory (KFD topology)
 //   croagent cxl    [--sysroot /]            CXL.mem inventory (/sys/bus/cxl)
 //   croagent pids   [--gpu-id N]             KFD compute processes
-//   croagent probe  [--device N]             gfx950 MFMA/HBM health probe
+//   croagent probe  [--device N | --bdf B]   gfx950 MFMA/HBM health probe
 //   croagent drain  --bdf 0000:5a:00.0       sysfs PCI remove
 //   croagent rescan                          sysfs PCI rescan
 //
@@ -257,6 +257,20 @@ int cmd_rescan() {
   return 0;
 }
 
+int resolve_device_by_bdf(const std::string& bdf) {
+  int count = cro_probe_device_count();
+  char buf[64];
+  std::string want = bdf.substr(0, bdf.rfind('.'));  // match domain:bus:dev
+  for (int dev = 0; dev < count; ++dev) {
+    if (cro_probe_pci_bus_id(dev, buf, sizeof(buf)) == 0) {
+      std::string got(buf);
+      for (auto& c : got) c = tolower(c);
+      if (got.rfind(want, 0) == 0) return dev;
+    }
+  }
+  return -1;
+}
+
 int cmd_probe(int device) {
   CroProbeResult result;
   int rc = cro_probe_run(device, &result);
@@ -290,7 +304,18 @@ int main(int argc, char** argv) {
   if (cmd == "list") return cmd_list();
   if (cmd == "cxl") return cmd_cxl();
   if (cmd == "pids") return cmd_pids(gpu_id);
-  if (cmd == "probe") return cmd_probe(device);
+  if (cmd == "probe") {
+    if (!bdf.empty()) {
+      std::string lower = bdf;
+      for (auto& c : lower) c = tolower(c);
+      device = resolve_device_by_bdf(lower);
+      if (device < 0) {
+        fprintf(stderr, "no HIP device with bdf %s\n", bdf.c_str());
+        return 1;
+      }
+    }
+    return cmd_probe(device);
+  }
   if (cmd == "drain") return cmd_drain(bdf);
   if (cmd == "rescan") return cmd_rescan();
   fprintf(stderr, "unknown command %s\n", cmd.c_str());

@@ -134,3 +134,30 @@ def probe_fn_for_nodeops(gpu) -> dict:
     if dev is None:
         dev = 0
     return run_probe(dev)
+
+
+def probe_via_exec(execer, node: str, gpu) -> dict:
+    """Health probe through the node agent (``croagent probe --bdf``) — the
+    cluster shape where controllers run off-node and reach hardware only
+    through the NodeExec seam.  Returns the same dict shape as run_probe.
+    """
+    import json as _json
+
+    rc, out, err = execer.run(
+        node, ["croagent", "probe", "--bdf", gpu.pci_bdf], timeout=300
+    )
+    if rc != 0 and not out.strip():
+        return {"ok": False, "rc": rc, "msg": err.strip() or "croagent probe failed"}
+    try:
+        return _json.loads(out)
+    except ValueError:
+        return {"ok": False, "rc": rc, "msg": f"unparseable probe output: {out[:200]}"}
+
+
+def make_exec_probe_fn(execer, node: str):
+    """AmdNodeOps probe hook bound to a NodeExec (local or remote agent)."""
+
+    def probe(gpu):
+        return probe_via_exec(execer, node, gpu)
+
+    return probe

@@ -138,3 +138,17 @@ def test_cxl_inventory(agent, tmp_path):
             "pci_bdf": "0000:60:00.0",
         }
     ]
+
+
+@pytest.mark.gpu
+def test_probe_by_bdf_on_real_gpu(agent):
+    if not os.path.exists("/dev/kfd"):
+        pytest.skip("no GPU")
+    from cro_amd.nodeops.execs import LocalNodeExec
+    from cro_amd.nodeops.kfd import enumerate_gpus
+
+    gpus = enumerate_gpus(LocalNodeExec(), "local")
+    proc = run_agent(agent, "probe", "--bdf", gpus[0].pci_bdf)
+    assert proc.returncode == 0, proc.stdout + proc.stderr
+    data = json.loads(proc.stdout)
+    assert data["ok"] and data["mfma_f32_exact"]

@@ -97,3 +97,39 @@ def test_agent_api_against_real_fs(tmp_path):
     remote.write_file("n", "/etc/cdi/spec.json", "{}")
     assert (tmp_path / "etc" / "cdi" / "spec.json").read_text() == "{}"
     assert remote.path_exists("n", "/sys/module/amdgpu")
+
+
+def test_probe_via_exec_parses_croagent_output(remote_exec):
+    """Exec-based probe hook: croagent JSON over the agent API."""
+    from cro_amd.nodeops.kfd import enumerate_gpus
+    from cro_amd.nodeops.probe import make_exec_probe_fn
+
+    remote, backend, ids = remote_exec
+    backend.set_command(
+        ("croagent", "probe", "--bdf", "0000:03:00.0"),
+        (0, '{"ok":true,"rc":0,"mfma_f32_exact":true,"hbm_gbps":4500.0,'
+            '"bf16_tflops":2000.0,"vram_total":309220868096,"vram_free":1,'
+            '"gcn_arch":"gfx950","msg":"ok"}', ""),
+    )
+    gpus = enumerate_gpus(remote, NODE)
+    probe = make_exec_probe_fn(remote, NODE)
+    result = probe(gpus[0])
+    assert result["ok"] and result["mfma_f32_exact"]
+    assert result["gcn_arch"] == "gfx950"
+
+
+def test_probe_via_exec_failure_shapes(remote_exec):
+    from cro_amd.nodeops.kfd import GPUDevice
+    from cro_amd.nodeops.probe import probe_via_exec
+
+    remote, backend, _ = remote_exec
+    gpu = GPUDevice(kfd_node=1, gpu_id=1, device_id="GPU-x", unique_id=1,
+                    render_minor=128, pci_bdf="0000:99:00.0")
+    backend.set_command(("croagent", "probe", "--bdf", "0000:99:00.0"),
+                        (1, "", "no HIP device with bdf 0000:99:00.0"))
+    result = probe_via_exec(remote, NODE, gpu)
+    assert not result["ok"] and "no HIP device" in result["msg"]
+    backend.set_command(("croagent", "probe", "--bdf", "0000:99:00.0"),
+                        (0, "garbage not json", ""))
+    result = probe_via_exec(remote, NODE, gpu)
+    assert not result["ok"] and "unparseable" in result["msg"]
