@@ -21,6 +21,7 @@ from cro_amd.api.v1alpha1.types import (
     Node,
 )
 from cro_amd.controllers.composabilityrequest import ComposabilityRequestReconciler
+from cro_amd.controllers.upstreamsyncer import UpstreamSyncer
 from cro_amd.controllers.composableresource import (
     ComposableResourceReconciler,
     ReconcileConfig,
@@ -46,6 +47,7 @@ class ControllerMachine(RuleBasedStateMachine):
             self.client, adapter, self.ops, ReconcileConfig()
         )
         self.req_rec = ComposabilityRequestReconciler(self.client)
+        self.syncer = UpstreamSyncer(self.client, adapter, self.ops, grace_period=0.0)
 
         orig_add = self.fabric.add_resource
 
@@ -114,6 +116,21 @@ class ControllerMachine(RuleBasedStateMachine):
         self.fabric.config.fail_attach = n
 
     @rule(node=st.sampled_from(["node0", "node1"]))
+    def fabric_drift(self, node):
+        # a device composes behind the operator's back (out-of-band)
+        free = [d for d in self.fabric._pool.values() if not d.attached_node]
+        if free:
+            self.fabric.force_attach(free[0].device_id, node)
+            self.ops.visible.setdefault(node, set()).add(free[0].device_id)
+
+    @rule()
+    def syncer_round(self):
+        try:
+            self.syncer.sync()
+        except Exception:
+            pass
+
+    @rule(node=st.sampled_from(["node0", "node1"]))
     def load_flips(self, node):
         if self.ops.loads.get(node):
             self.ops.clear_loads(node)
@@ -129,22 +146,19 @@ class ControllerMachine(RuleBasedStateMachine):
                 assert res.status.device_id, res
 
     @invariant()
-    def attached_devices_have_owners_or_pending(self):
-        # every fabric attachment is claimed by some CR's device_id OR a CR
-        # is still mid-attach on that node (identity write pending)
-        claimed = {
-            r.status.device_id
-            for r in self.client.list(ComposableResource)
-            if r.status.device_id
-        }
-        attaching_nodes = {
-            r.spec.target_node
-            for r in self.client.list(ComposableResource)
-            if r.status.state in ("", "Attaching")
-        }
-        for node in ("node0", "node1"):
-            for did in self.fabric.attached_to(node):
-                assert did in claimed or node in attaching_nodes, (did, node)
+    def no_phantom_claims(self):
+        """Every device an ONLINE CR claims is genuinely attached on the
+        fabric (the converse — unclaimed fabric attachments — is legal
+        drift until the syncer repairs it, which teardown proves)."""
+        attached = set(self.fabric.attached_to("node0")) | set(
+            self.fabric.attached_to("node1")
+        )
+        for r in self.client.list(ComposableResource):
+            if r.status.state == "Online" and r.status.device_id:
+                assert r.status.device_id in attached, (
+                    r.metadata.name,
+                    r.status.device_id,
+                )
 
     def teardown(self):
         # quiesce: heal everything, delete everything, reconcile to drain
@@ -157,11 +171,20 @@ class ControllerMachine(RuleBasedStateMachine):
                 self.client.delete(ComposabilityRequest, req.metadata.name)
             except Exception:
                 pass
-        for _ in range(40):
+        import time as _time
+
+        for _ in range(60):
+            try:
+                self.syncer.sync()  # grace 0: drift → detach CRs
+            except Exception:
+                pass
+            _time.sleep(0.001)
             self._reconcile_all_once()
             if (
                 not self.client.list(ComposabilityRequest)
                 and not self.client.list(ComposableResource)
+                and self.fabric.attached_to("node0") == []
+                and self.fabric.attached_to("node1") == []
             ):
                 break
         assert self.client.list(ComposabilityRequest) == []
