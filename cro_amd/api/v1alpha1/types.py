@@ -242,6 +242,27 @@ class DaemonSet(K8sObject):
     status: DaemonSetStatus = Field(default_factory=DaemonSetStatus)
 
 
+class Event(K8sObject):
+    """Lifecycle event stream (beyond the reference, which emits no
+    Kubernetes Events anywhere — no EventRecorder in any controller).
+    Flat core/v1-Event-shaped record: who it is about, what happened, and
+    a dedup count bumped on repeats. In cluster mode these map 1:1 onto
+    corev1 Events; in standalone mode they are served by the REST API and
+    read with ``croctl events``."""
+
+    KIND: ClassVar[str] = "Event"
+    apiVersion: str = "v1"
+    involved_kind: str = ""
+    involved_name: str = ""
+    type: str = "Normal"  # Normal | Warning
+    reason: str = ""
+    message: str = ""
+    count: int = 1
+    first_seen: str = ""
+    last_seen: str = ""
+    source: str = "cro-amd"
+
+
 ALL_KINDS = {
     cls.KIND: cls
     for cls in (
@@ -251,5 +272,6 @@ ALL_KINDS = {
         ResourceSlice,
         DeviceTaintRule,
         DaemonSet,
+        Event,
     )
 }

@@ -6,6 +6,7 @@
     croctl apply -f request.yaml
     croctl delete composabilityrequests NAME
     croctl scale composabilityrequests NAME --size N
+    croctl events [--for KIND/NAME]          # lifecycle audit trail
 
 Server selection: --server or CRO_SERVER (default http://127.0.0.1:8080).
 """
@@ -52,6 +53,14 @@ _COLUMNS["resourceslices"] = (
     ("DRIVER", lambda o: o["spec"]["driver"]),
     ("DEVICES", lambda o: str(len(o["spec"]["devices"]))),
 )
+_COLUMNS["events"] = (
+    ("LAST-SEEN", lambda o: (o.get("last_seen", "") or "")[11:19]),
+    ("TYPE", lambda o: o.get("type", "")),
+    ("REASON", lambda o: o.get("reason", "")),
+    ("OBJECT", lambda o: f'{o.get("involved_kind", "")}/{o.get("involved_name", "")}'),
+    ("COUNT", lambda o: str(o.get("count", 1))),
+    ("MESSAGE", lambda o: (o.get("message", "") or "")[:60]),
+)
 _DEFAULT_COLUMNS = (
     ("NAME", lambda o: o["metadata"]["name"]),
     ("KIND", lambda o: o.get("kind", "")),
@@ -97,6 +106,10 @@ def main(argv=None, client: httpx.Client = None) -> int:
     sc.add_argument("name")
     sc.add_argument("--size", type=int, required=True)
 
+    ev = sub.add_parser("events")
+    ev.add_argument("--for", dest="for_object", default="",
+                    metavar="KIND/NAME", help="filter by involved object")
+
     args = p.parse_args(argv)
     http = client or httpx.Client(base_url=args.server, timeout=30)
 
@@ -121,6 +134,26 @@ def main(argv=None, client: httpx.Client = None) -> int:
             print(yaml.safe_dump(items, sort_keys=False))
         else:
             _print_table(args.plural, items)
+        return 0
+
+    if args.command == "events":
+        resp = http.get(f"{BASE}/events")
+        if resp.status_code != 200:
+            return fail(resp)
+        items = sorted(resp.json()["items"], key=lambda o: o.get("last_seen", ""))
+        if args.for_object:
+            kind, _, name = args.for_object.partition("/")
+            items = [
+                o for o in items
+                if o.get("involved_kind", "").lower() == kind.lower()
+                and (not name or o.get("involved_name", "") == name)
+            ]
+        if args.output == "json":
+            print(json.dumps(items, indent=2))
+        elif args.output == "yaml":
+            print(yaml.safe_dump(items, sort_keys=False))
+        else:
+            _print_table("events", items)
         return 0
 
     if args.command == "describe":

@@ -54,9 +54,17 @@ class RequestReconcileConfig:
 
 
 class ComposabilityRequestReconciler(Reconciler):
-    def __init__(self, client: Client, config: Optional[RequestReconcileConfig] = None):
+    def __init__(
+        self,
+        client: Client,
+        config: Optional[RequestReconcileConfig] = None,
+        recorder=None,
+    ):
+        from ..runtime.events import NullRecorder
+
         self.client = client
         self.config = config or RequestReconcileConfig()
+        self.recorder = recorder or NullRecorder()
 
     # -- plumbing ----------------------------------------------------------
 
@@ -99,6 +107,7 @@ class ComposabilityRequestReconciler(Reconciler):
             self.client.update_status(fresh)
         except Exception:
             pass
+        self.recorder.warning(request, "ReconcileError", msg)
 
     def _garbage_collect(self, request: ComposabilityRequest) -> bool:
         """Target node deleted → delete the request (:147-167)."""
@@ -278,6 +287,12 @@ class ComposabilityRequestReconciler(Reconciler):
         request.status.error = ""
         request.status.scalarResource = spec.model_copy(deep=True)
         self.client.update_status(request)
+        nodes_used = sorted({e.node_name for e in request.status.resources.values()})
+        self.recorder.normal(
+            request,
+            "NodesAllocated",
+            f"{len(request.status.resources)} device(s) on {', '.join(nodes_used)}",
+        )
         return Result()
 
     def _node_occupied(self, node_name: str, request, all_requests) -> bool:
@@ -336,6 +351,11 @@ class ComposabilityRequestReconciler(Reconciler):
             request.status.error = ""
             request.status.scalarResource = request.spec.resource.model_copy(deep=True)
             self.client.update_status(request)
+            self.recorder.normal(
+                request,
+                "Running",
+                f"all {len(request.status.resources)} device(s) online",
+            )
             return Result()
         return Result(requeue_after=self.config.updating_wait)
 
@@ -350,6 +370,9 @@ class ComposabilityRequestReconciler(Reconciler):
             request.status.state = "NodeAllocating"
             request.status.scalarResource = request.spec.resource.model_copy(deep=True)
             self.client.update_status(request)
+            self.recorder.normal(
+                request, "SpecChanged", "spec drift detected; re-allocating"
+            )
             return Result()
 
         if request.status.error:

@@ -68,11 +68,15 @@ class ComposableResourceReconciler(Reconciler):
         adapter: Adapter,
         node_ops: amdgpu.NodeOps,
         config: Optional[ReconcileConfig] = None,
+        recorder=None,
     ):
+        from ..runtime.events import NullRecorder
+
         self.client = client
         self.adapter = adapter
         self.node_ops = node_ops
         self.config = config or ReconcileConfig()
+        self.recorder = recorder or NullRecorder()
         self.metrics = Metrics()
         # uid → monotonic time of Attaching entry (restart loses the sample,
         # never corrupts it — the histogram only sees fully observed attaches)
@@ -191,6 +195,7 @@ class ComposableResourceReconciler(Reconciler):
             self.client.update_status(fresh)
         except Exception:
             log.debug("could not record error on %s", resource.metadata.name)
+        self.recorder.warning(resource, "ReconcileError", msg)
 
     def _garbage_collect(self, resource: ComposableResource) -> bool:
         """Target node deleted → taint cleanup, Deleting, delete CR
@@ -212,6 +217,12 @@ class ComposableResourceReconciler(Reconciler):
         if resource.metadata.deletionTimestamp is None:
             self.client.delete(resource)
             did = True
+        if did:
+            self.recorder.warning(
+                resource,
+                "GarbageCollected",
+                f"target node {resource.spec.target_node} deleted",
+            )
         return did
 
     # -- states ------------------------------------------------------------
@@ -234,6 +245,12 @@ class ComposableResourceReconciler(Reconciler):
         resource.status.state = "Attaching"
         resource.status.error = ""
         self.client.update_status(resource)
+        self.recorder.normal(
+            resource,
+            "AttachStarted",
+            f"attaching {resource.spec.type}/{resource.spec.model} "
+            f"on {resource.spec.target_node}",
+        )
         return Result()
 
     def _handle_attaching(self, resource: ComposableResource) -> Result:
@@ -274,6 +291,9 @@ class ComposableResourceReconciler(Reconciler):
                 resource.metadata.name, device_id, cdi_device_id
             )
             self._fabric_polls.pop(resource.metadata.uid, None)
+            self.recorder.normal(
+                resource, "FabricAttached", f"fabric composed device {device_id}"
+            )
 
         if mode == "DEVICE_PLUGIN":
             # load check is advisory on attach (reference logs and continues,
@@ -308,6 +328,17 @@ class ComposableResourceReconciler(Reconciler):
         resource.status.state = "Online"
         resource.status.error = ""
         self.client.update_status(resource)
+        probe_note = ""
+        if probe is not None:
+            probe_note = (
+                f"; probe: mfma_exact={probe.get('mfma_f32_exact')} "
+                f"hbm={probe.get('hbm_gbps', 0):.0f} GB/s"
+            )
+        self.recorder.normal(
+            resource,
+            "Online",
+            f"device {resource.status.device_id} online with CDI spec{probe_note}",
+        )
         started = self._attach_started.pop(resource.metadata.uid, None)
         if started is not None:
             self.metrics.attach_to_ready_seconds.observe(time.monotonic() - started)
@@ -320,6 +351,11 @@ class ComposableResourceReconciler(Reconciler):
             resource.status.state = "Detaching"
             self.client.update_status(resource)
             self.metrics.devices_online.dec()
+            self.recorder.normal(
+                resource,
+                "DetachStarted",
+                f"detaching device {resource.status.device_id}",
+            )
             return Result()
 
         if resource.metadata.labels.get(READY_TO_DETACH_LABEL, ""):
@@ -334,6 +370,7 @@ class ComposableResourceReconciler(Reconciler):
         except FabricError as exc:
             resource.status.error = str(exc)
             self.client.update_status(resource)
+            self.recorder.warning(resource, "HealthCheckFailed", str(exc))
             return Result(requeue_after=self.config.online_health_period)
         finally:
             self.metrics.fabric_request_seconds.labels(
@@ -401,6 +438,7 @@ class ComposableResourceReconciler(Reconciler):
             started = self._detach_started.pop(resource.metadata.uid, None)
             if started is not None:
                 self.metrics.detach_seconds.observe(time.monotonic() - started)
+            self.recorder.normal(resource, "Detached", "device returned to fabric pool")
 
         resource.status.state = "Deleting"
         self.client.update_status(resource)

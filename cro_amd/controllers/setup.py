@@ -59,13 +59,19 @@ def build_manager(
     syncer_grace: float = 600.0,
     metrics_port: Optional[int] = None,
     client=None,
+    record_events: bool = True,
 ) -> Manager:
+    from ..runtime.events import EventRecorder, NullRecorder
+
     mgr = Manager(store=store, metrics_port=metrics_port, client=client)
+    recorder = EventRecorder(mgr.client) if record_events else NullRecorder()
 
     resource_reconciler = ComposableResourceReconciler(
-        mgr.client, adapter, node_ops, resource_config
+        mgr.client, adapter, node_ops, resource_config, recorder=recorder
     )
-    request_reconciler = ComposabilityRequestReconciler(mgr.client, request_config)
+    request_reconciler = ComposabilityRequestReconciler(
+        mgr.client, request_config, recorder=recorder
+    )
 
     mgr.add_controller(
         Controller(
@@ -95,7 +101,10 @@ def build_manager(
         mgr.register_admission("ComposabilityRequest", admission_validator(mgr.client))
 
     if syncer_period is not None:
-        syncer = UpstreamSyncer(mgr.client, adapter, node_ops, grace_period=syncer_grace)
+        syncer = UpstreamSyncer(
+            mgr.client, adapter, node_ops,
+            grace_period=syncer_grace, recorder=recorder,
+        )
         mgr.add_runnable(syncer_period, syncer.sync)
         mgr.syncer = syncer  # exposed for tests
 

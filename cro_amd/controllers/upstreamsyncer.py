@@ -34,11 +34,15 @@ class UpstreamSyncer:
         adapter: Adapter,
         node_ops: NodeOps,
         grace_period: float = DEFAULT_GRACE_PERIOD,
+        recorder=None,
     ):
+        from ..runtime.events import NullRecorder
+
         self.client = client
         self.adapter = adapter
         self.node_ops = node_ops
         self.grace_period = grace_period
+        self.recorder = recorder or NullRecorder()
         self.missing_devices: Dict[str, float] = {}
 
     def sync(self) -> None:
@@ -61,6 +65,11 @@ class UpstreamSyncer:
             if first_seen is None:
                 log.info("upstream device %s has no local CR; tracking with grace", did)
                 self.missing_devices[did] = time.monotonic()
+                self.recorder.warning(
+                    ("Node", info.node_name),
+                    "FabricDrift",
+                    f"device {did} attached upstream with no local CR",
+                )
             elif time.monotonic() - first_seen > self.grace_period:
                 log.info("grace exceeded for %s; creating detach CR", did)
                 try:
@@ -89,3 +98,8 @@ class UpstreamSyncer:
         cr.metadata.labels[READY_TO_DETACH_LABEL] = info.device_id
         cr.metadata.labels[READY_TO_DETACH_CDI_LABEL] = info.cdi_device_id
         self.client.create(cr)
+        self.recorder.normal(
+            ("Node", info.node_name),
+            "DriftDetachCreated",
+            f"detach CR created for unclaimed device {info.device_id}",
+        )

@@ -1,0 +1,126 @@
+"""Event recorder: the controllers' audit trail.
+
+The reference emits no Kubernetes Events at all (no EventRecorder is
+constructed anywhere under internal/controller — verified by grep); the
+only way to see why an attach stalled is operator logs. This build gives
+every lifecycle transition a durable, queryable Event object:
+
+* content-addressed dedup as in core/v1 Events: repeats of the same
+  (object, reason, message) bump ``count``/``last_seen`` instead of
+  piling up new objects;
+* bounded retention (oldest-by-last_seen eviction past ``max_events``)
+  so churn cannot grow the store without limit;
+* never throws: observability must not break a reconcile. Conflicts
+  (two workers recording the same event) retry once, then drop.
+
+Thread-safe to the extent the underlying store is (all mutations go
+through the client's optimistic-concurrency writes).
+"""
+
+from __future__ import annotations
+
+import hashlib
+import logging
+from datetime import datetime, timezone
+from typing import Optional
+
+from ..api.v1alpha1.types import Event
+from .errors import AlreadyExistsError, ConflictError, NotFoundError
+
+log = logging.getLogger("cro.events")
+
+NORMAL = "Normal"
+WARNING = "Warning"
+
+
+def _now() -> str:
+    return datetime.now(timezone.utc).isoformat()
+
+
+class EventRecorder:
+    #: eviction runs every N creates, not on each one — listing the whole
+    #: event set deep-copies it, which must stay off the hot attach path
+    #: (the attach p50 is ~4.5 ms; an O(max_events) scan per event would
+    #: be a measurable fraction of that). Worst-case overshoot is N.
+    EVICT_EVERY = 64
+
+    def __init__(self, client, max_events: int = 1000, source: str = "cro-amd"):
+        self.client = client
+        self.max_events = max_events
+        self.source = source
+        self._creates = 0
+
+    # -- public API --------------------------------------------------------
+
+    def event(self, obj, type_: str, reason: str, message: str) -> None:
+        """Record one event about ``obj`` (a K8sObject, or a
+        ``(kind, name)`` tuple). Never raises."""
+        try:
+            self._record(obj, type_, reason, message)
+        except Exception:  # pragma: no cover - defensive
+            log.exception("event recording failed (%s/%s)", reason, message)
+
+    def normal(self, obj, reason: str, message: str) -> None:
+        self.event(obj, NORMAL, reason, message)
+
+    def warning(self, obj, reason: str, message: str) -> None:
+        self.event(obj, WARNING, reason, message)
+
+    # -- internals ---------------------------------------------------------
+
+    def _record(self, obj, type_: str, reason: str, message: str) -> None:
+        if isinstance(obj, tuple):
+            kind, name = obj
+        else:
+            kind, name = obj.KIND, obj.metadata.name
+        digest = hashlib.sha256(
+            f"{kind}/{name}/{type_}/{reason}/{message}".encode()
+        ).hexdigest()[:12]
+        ev_name = f"{name}.{digest}".lower()
+
+        for _ in range(2):  # one conflict retry
+            existing = self.client.try_get(Event, ev_name)
+            try:
+                if existing is None:
+                    ev = Event()
+                    ev.metadata.name = ev_name
+                    ev.involved_kind = kind
+                    ev.involved_name = name
+                    ev.type = type_
+                    ev.reason = reason
+                    ev.message = message
+                    ev.count = 1
+                    ev.first_seen = ev.last_seen = _now()
+                    ev.source = self.source
+                    self.client.create(ev)
+                    self._creates += 1
+                    if self._creates % self.EVICT_EVERY == 0:
+                        self._evict_over_cap()
+                else:
+                    existing.count += 1
+                    existing.last_seen = _now()
+                    self.client.update(existing)
+                return
+            except (ConflictError, AlreadyExistsError, NotFoundError):
+                continue  # lost a race with a sibling worker; re-read once
+
+    def _evict_over_cap(self) -> None:
+        events = self.client.list(Event)
+        excess = len(events) - self.max_events
+        if excess <= 0:
+            return
+        for ev in sorted(events, key=lambda e: e.last_seen)[:excess]:
+            try:
+                self.client.delete(Event, ev.metadata.name)
+            except Exception:
+                pass
+
+
+class NullRecorder(EventRecorder):
+    """Default when no recorder is wired: everything is a no-op."""
+
+    def __init__(self):  # noqa: D401 - intentionally no client
+        pass
+
+    def event(self, obj, type_, reason, message):
+        return None

@@ -41,6 +41,7 @@ _PLURALS = {cls.KIND: plural for plural, cls in {
     "resourceslices": ALL_KINDS["ResourceSlice"],
     "devicetaintrules": ALL_KINDS["DeviceTaintRule"],
     "nodes": ALL_KINDS["Node"],
+    "events": ALL_KINDS["Event"],
 }.items()}
 
 

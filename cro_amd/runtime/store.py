@@ -177,9 +177,13 @@ class InMemoryStore:
                     "deletionTimestamp": stored.metadata.deletionTimestamp,
                 }
             )
-            if incoming_meta == stored.metadata and getattr(obj, "spec", None) == getattr(
-                stored, "spec", None
-            ):
+            # compare the WHOLE object minus status (not just spec): kinds
+            # with top-level fields (Event count/last_seen) must not be
+            # misread as no-ops
+            incoming_dump = obj.model_copy(
+                update={"metadata": incoming_meta}
+            ).model_dump(exclude={"status"})
+            if incoming_dump == stored.model_dump(exclude={"status"}):
                 return stored.model_copy(deep=True)
             self._admit("UPDATE", stored, obj)
             new = obj

@@ -24,6 +24,7 @@ from ..api.v1alpha1.types import (
     ComposabilityRequest,
     ComposableResource,
     DeviceTaintRule,
+    Event,
     Node,
     ResourceSlice,
 )
@@ -41,6 +42,7 @@ PLURALS = {
     "resourceslices": ResourceSlice,
     "devicetaintrules": DeviceTaintRule,
     "nodes": Node,
+    "events": Event,
 }
 
 BASE = "/apis/cro.hpsys.ibm.ie.com/v1alpha1"

@@ -104,13 +104,18 @@ def mock_world(client):
     class World:
         pass
 
+    from cro_amd.runtime.events import EventRecorder
+
     w = World()
     w.client = client
     w.fabric = fabric
     w.adapter = adapter
     w.ops = ops
-    w.resource_rec = ComposableResourceReconciler(client, adapter, ops, ReconcileConfig())
-    w.request_rec = ComposabilityRequestReconciler(client)
+    w.recorder = EventRecorder(client)
+    w.resource_rec = ComposableResourceReconciler(
+        client, adapter, ops, ReconcileConfig(), recorder=w.recorder
+    )
+    w.request_rec = ComposabilityRequestReconciler(client, recorder=w.recorder)
     return w
 
 
