@@ -110,6 +110,11 @@ def main(argv=None, client: httpx.Client = None) -> int:
     ev.add_argument("--for", dest="for_object", default="",
                     metavar="KIND/NAME", help="filter by involved object")
 
+    wa = sub.add_parser("watch")
+    wa.add_argument("plural")
+    wa.add_argument("--count", type=int, default=0,
+                    help="exit after N events (0 = forever)")
+
     args = p.parse_args(argv)
     http = client or httpx.Client(base_url=args.server, timeout=30)
 
@@ -134,6 +139,25 @@ def main(argv=None, client: httpx.Client = None) -> int:
             print(yaml.safe_dump(items, sort_keys=False))
         else:
             _print_table(args.plural, items)
+        return 0
+
+    if args.command == "watch":
+        # k8s-style list+watch stream: synthetic ADDED replay, then live
+        seen = 0
+        with http.stream("GET", f"{BASE}/{args.plural}?watch=true") as resp:
+            if resp.status_code != 200:
+                print(f"error: {resp.status_code}", file=sys.stderr)
+                return 1
+            for line in resp.iter_lines():
+                if not line.strip():
+                    continue  # keepalive
+                ev = json.loads(line)
+                obj = ev["object"]
+                state = (obj.get("status") or {}).get("state", "")
+                print(f'{ev["type"]:<9} {obj["metadata"]["name"]} {state}')
+                seen += 1
+                if args.count and seen >= args.count:
+                    return 0
         return 0
 
     if args.command == "events":

@@ -100,3 +100,16 @@ def test_error_paths(cli):
     assert rc == 1 and "404" in err
     rc, _, err = run("delete", "composabilityrequests", "nope")
     assert rc == 1
+
+
+def test_events_after_lifecycle(cli, tmp_path):
+    run, stack = cli
+    run("apply", "-f", write_request(tmp_path, name="e1"))
+    assert stack.mgr.wait_for(
+        lambda: "Running" in run("get", "composabilityrequests", "e1")[1],
+        timeout=10,
+    )
+    rc, out, _ = run("events")
+    assert rc == 0 and "NodesAllocated" in out and "Online" in out
+    rc, out, _ = run("events", "--for", "ComposabilityRequest/e1")
+    assert rc == 0 and "Running" in out

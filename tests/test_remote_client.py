@@ -233,3 +233,21 @@ def test_api_server_outage_mid_lifecycle_recovers():
         remote.close()
         server.should_exit = True
         thread.join(timeout=5)
+
+
+def test_croctl_watch_over_http(api_server, capsys):
+    """croctl watch consumes the ndjson list+watch stream over a real
+    socket (the ASGI test client cannot terminate an infinite stream, so
+    this lives here with the uvicorn fixture)."""
+    from cro_amd.cmd.croctl import main as croctl
+
+    url, server_mgr = api_server
+    server_mgr.client.create(make_request("w1", target_node="node0"))
+    server_mgr.client.create(make_request("w2", model="mi300x", target_node="node0"))
+    rc = croctl(["--server", url, "watch", "composabilityrequests", "--count", "2"])
+    out = capsys.readouterr().out
+    assert rc == 0
+    lines = [ln for ln in out.splitlines() if ln.strip()]
+    assert len(lines) == 2
+    assert all(ln.startswith("ADDED") for ln in lines)
+    assert {"w1", "w2"} <= {ln.split()[1] for ln in lines}
