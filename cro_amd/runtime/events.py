@@ -22,7 +22,6 @@ from __future__ import annotations
 import hashlib
 import logging
 from datetime import datetime, timezone
-from typing import Optional
 
 from ..api.v1alpha1.types import Event
 from .errors import AlreadyExistsError, ConflictError, NotFoundError
