@@ -62,6 +62,7 @@ def build_local_stack(
     max_concurrent_reconciles: int = 8,
     syncer_period: Optional[float] = 1.0,
     syncer_grace: float = 10.0,
+    record_events: bool = True,
 ) -> LocalStack:
     if use_gpu is None:
         use_gpu = gpu_available()
@@ -99,6 +100,7 @@ def build_local_stack(
             max_concurrent_reconciles=max_concurrent_reconciles,
             syncer_period=syncer_period,
             syncer_grace=syncer_grace,
+            record_events=record_events,
         )
         ops = AmdNodeOps(
             execer,
@@ -134,6 +136,7 @@ def build_local_stack(
             max_concurrent_reconciles=max_concurrent_reconciles,
             syncer_period=syncer_period,
             syncer_grace=syncer_grace,
+            record_events=record_events,
         )
         ops = MockNodeOps(client=mgr.client)
         mgr.resource_reconciler.node_ops = ops

@@ -64,7 +64,12 @@ def build_manager(
     from ..runtime.events import EventRecorder, NullRecorder
 
     mgr = Manager(store=store, metrics_port=metrics_port, client=client)
-    recorder = EventRecorder(mgr.client) if record_events else NullRecorder()
+    # buffered (k8s EventBroadcaster shape): reconciles only enqueue;
+    # a daemon thread does the store writes off the hot path
+    recorder = (
+        EventRecorder(mgr.client, asynchronous=True) if record_events else NullRecorder()
+    )
+    mgr.recorder = recorder
 
     resource_reconciler = ComposableResourceReconciler(
         mgr.client, adapter, node_ops, resource_config, recorder=recorder

@@ -109,6 +109,7 @@ def test_events_after_lifecycle(cli, tmp_path):
         lambda: "Running" in run("get", "composabilityrequests", "e1")[1],
         timeout=10,
     )
+    stack.mgr.recorder.flush()  # buffered recorder: drain before asserting
     rc, out, _ = run("events")
     assert rc == 0 and "NodesAllocated" in out and "Online" in out
     rc, out, _ = run("events", "--for", "ComposabilityRequest/e1")

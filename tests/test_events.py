@@ -54,6 +54,46 @@ def test_retention_evicts_oldest(client):
     assert "obj-24" in names  # newest kept
 
 
+def test_async_mode_drains_off_thread(client):
+    rec = EventRecorder(client, asynchronous=True)
+    for i in range(20):
+        rec.normal(("X", f"o{i}"), "R", "m")
+    assert rec.flush(timeout=10)
+    assert len(client.list(Event)) == 20
+    # dedup still applies through the buffer
+    for _ in range(3):
+        rec.normal(("X", "o0"), "R", "m")
+    assert rec.flush(timeout=10)
+    ev = [e for e in client.list(Event) if e.involved_name == "o0"][0]
+    assert ev.count == 4
+
+
+def test_async_overflow_drops_oldest(client):
+    class SlowClient:
+        """Stand-in that blocks writes so the buffer can overflow."""
+
+        def __init__(self, inner):
+            self.inner = inner
+            self.gate = __import__("threading").Event()
+
+        def try_get(self, *a):
+            self.gate.wait(10)
+            return self.inner.try_get(*a)
+
+        def __getattr__(self, name):
+            return getattr(self.inner, name)
+
+    slow = SlowClient(client)
+    rec = EventRecorder(slow, asynchronous=True, buffer_size=8)
+    for i in range(50):
+        rec.normal(("X", f"o{i}"), "R", "m")
+    slow.gate.set()
+    assert rec.flush(timeout=10)
+    stored = {e.involved_name for e in client.list(Event)}
+    assert len(stored) <= 9  # bounded: 8 buffered + ≤1 in flight
+    assert "o49" in stored  # newest survived the overflow
+
+
 def test_recorder_never_raises():
     class Broken:
         def try_get(self, *a):
