@@ -98,12 +98,19 @@ class EventRecorder:
 
         deadline = _time.monotonic() + timeout
         with self._wake:
-            self._wake.notify()
             while self._buf or self._inflight:
+                # the worker idles out after 30 s; an event appended in that
+                # exit window needs the worker restarted, not just notified
+                if self._buf and (self._worker is None or not self._worker.is_alive()):
+                    self._worker = threading.Thread(
+                        target=self._drain_loop, name="event-recorder", daemon=True
+                    )
+                    self._worker.start()
+                self._wake.notify_all()
                 remaining = deadline - _time.monotonic()
                 if remaining <= 0:
                     return False
-                self._wake.wait(remaining)
+                self._wake.wait(min(remaining, 0.1))
         return True
 
     def _drain_loop(self) -> None:
