@@ -60,7 +60,13 @@ def main(argv=None) -> int:
     p.add_argument("--syncer-period", type=float, default=60.0)
     p.add_argument("--zap-log-level", default="info")
     p.add_argument("--log-format", choices=["text", "json"], default="text")
+    p.add_argument("--tls-cert-file", default="",
+                   help="serve the API over TLS (the reference serves its "
+                   "webhook and metrics over TLS from cert-manager certs)")
+    p.add_argument("--tls-key-file", default="")
     args = p.parse_args(argv)
+    if bool(args.tls_cert_file) != bool(args.tls_key_file):
+        p.error("--tls-cert-file and --tls-key-file must be given together")
 
     if args.log_format == "json":
         import json as _json
@@ -229,6 +235,8 @@ def main(argv=None) -> int:
             host="0.0.0.0",
             port=parse_port(args.api_bind_address, 8080),
             log_level="warning",
+            ssl_certfile=args.tls_cert_file or None,
+            ssl_keyfile=args.tls_key_file or None,
         )
     )
 

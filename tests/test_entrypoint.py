@@ -117,3 +117,60 @@ def test_leader_election_flock(tmp_path):
         fcntl.flock(second, fcntl.LOCK_EX | fcntl.LOCK_NB)
     fcntl.flock(first, fcntl.LOCK_UN)
     fcntl.flock(second, fcntl.LOCK_EX | fcntl.LOCK_NB)  # now acquirable
+
+
+@pytest.mark.timeout(120)
+def test_main_serves_tls(tmp_path):
+    """--tls-cert-file/--tls-key-file serve the API over TLS (the
+    reference's webhook/metrics are TLS-only behind cert-manager certs,
+    cmd/main.go:109-127 + config/certmanager)."""
+    cert = tmp_path / "tls.crt"
+    key = tmp_path / "tls.key"
+    subprocess.run(
+        ["openssl", "req", "-x509", "-newkey", "rsa:2048", "-nodes",
+         "-keyout", str(key), "-out", str(cert), "-days", "1",
+         "-subj", "/CN=127.0.0.1"],
+        check=True, capture_output=True,
+    )
+    port = free_port()
+    env = dict(os.environ)
+    env.update({"DEVICE_RESOURCE_TYPE": "DRA", "CDI_PROVIDER_TYPE": "MOCK"})
+    proc = subprocess.Popen(
+        [
+            sys.executable, "-m", "cro_amd.cmd.main",
+            "--api-bind-address", f":{port}",
+            "--metrics-bind-address", f":{free_port()}",
+            "--serve-only",
+            "--tls-cert-file", str(cert), "--tls-key-file", str(key),
+        ],
+        cwd=REPO, env=env,
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+    )
+    try:
+        base = f"https://127.0.0.1:{port}"
+        deadline = time.monotonic() + 30
+        up = False
+        while time.monotonic() < deadline:
+            try:
+                if httpx.get(base + "/healthz", timeout=1, verify=False).status_code == 200:
+                    up = True
+                    break
+            except Exception:
+                time.sleep(0.2)
+        assert up, proc.stdout.read() if proc.poll() is not None else "no TLS healthz"
+        # plain HTTP against the TLS port must fail
+        with pytest.raises(Exception):
+            httpx.get(f"http://127.0.0.1:{port}/healthz", timeout=2)
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+
+
+def test_tls_flags_must_pair():
+    from cro_amd.cmd.main import main as entry
+
+    with pytest.raises(SystemExit):
+        entry(["--tls-cert-file", "/tmp/only-cert.pem"])
