@@ -84,6 +84,10 @@ def main(argv=None, client: httpx.Client = None) -> int:
     p = argparse.ArgumentParser(prog="croctl")
     p.add_argument("--server", default=os.environ.get("CRO_SERVER", "http://127.0.0.1:8080"))
     p.add_argument("-o", "--output", choices=["table", "yaml", "json"], default="table")
+    p.add_argument("--insecure-skip-tls-verify", action="store_true",
+                   help="accept any server certificate (self-signed demos)")
+    p.add_argument("--certificate-authority", default="",
+                   metavar="CA_PEM", help="CA bundle for the server certificate")
     sub = p.add_subparsers(dest="command", required=True)
 
     g = sub.add_parser("get")
@@ -116,7 +120,12 @@ def main(argv=None, client: httpx.Client = None) -> int:
                     help="exit after N events (0 = forever)")
 
     args = p.parse_args(argv)
-    http = client or httpx.Client(base_url=args.server, timeout=30)
+    verify = True
+    if args.insecure_skip_tls_verify:
+        verify = False
+    elif args.certificate_authority:
+        verify = args.certificate_authority
+    http = client or httpx.Client(base_url=args.server, timeout=30, verify=verify)
 
     def fail(resp) -> int:
         print(f"error: {resp.status_code}: {resp.text}", file=sys.stderr)

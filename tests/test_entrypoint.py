@@ -129,7 +129,8 @@ def test_main_serves_tls(tmp_path):
     subprocess.run(
         ["openssl", "req", "-x509", "-newkey", "rsa:2048", "-nodes",
          "-keyout", str(key), "-out", str(cert), "-days", "1",
-         "-subj", "/CN=127.0.0.1"],
+         "-subj", "/CN=127.0.0.1",
+         "-addext", "subjectAltName=IP:127.0.0.1"],
         check=True, capture_output=True,
     )
     port = free_port()
@@ -161,6 +162,13 @@ def test_main_serves_tls(tmp_path):
         # plain HTTP against the TLS port must fail
         with pytest.raises(Exception):
             httpx.get(f"http://127.0.0.1:{port}/healthz", timeout=2)
+        # croctl reaches it with --insecure-skip-tls-verify or the CA
+        from cro_amd.cmd.croctl import main as croctl
+
+        assert croctl(["--server", base, "--insecure-skip-tls-verify",
+                       "get", "nodes"]) == 0
+        assert croctl(["--server", base, "--certificate-authority", str(cert),
+                       "get", "nodes"]) == 0
     finally:
         proc.send_signal(signal.SIGTERM)
         try:
