@@ -21,9 +21,19 @@ Resolver = Union[Dict[str, str], Callable[[str], str]]
 
 
 class RemoteNodeExec(NodeExec):
-    def __init__(self, resolver: Resolver, transport: Optional[httpx.BaseTransport] = None):
+    def __init__(
+        self,
+        resolver: Resolver,
+        transport: Optional[httpx.BaseTransport] = None,
+        token: Optional[str] = None,
+    ):
+        import os
+
         self._resolver = resolver
-        self._http = httpx.Client(transport=transport, timeout=90)
+        if token is None:
+            token = os.environ.get("CRO_AGENT_TOKEN", "")
+        headers = {"Authorization": f"Bearer {token}"} if token else None
+        self._http = httpx.Client(transport=transport, timeout=90, headers=headers)
 
     def _url(self, node: str, path: str) -> str:
         if callable(self._resolver):

@@ -13,9 +13,18 @@ API (cro_amd/runtime exposes the matching client, RemoteNodeExec):
 
 Only the NodeExec verbs are exposed — no arbitrary shell.  Binary execution
 inherits LocalNodeExec's trusted-path resolution (never $PATH).
+
+Authentication: set ``CRO_AGENT_TOKEN`` (or pass ``token=``) and every
+/agent route requires ``Authorization: Bearer <token>`` — the standalone
+analog of the RBAC that gated the reference's pods/exec path
+(config/rbac/role.yaml pods/exec verbs). RemoteNodeExec sends the token
+from the same env var.
 """
 
 from __future__ import annotations
+
+import hmac
+import os
 
 from fastapi import FastAPI, HTTPException, Request, Response
 
@@ -23,10 +32,21 @@ from ..nodeops.execs import ExecError, NodeExec
 
 
 def build_agent_app(
-    execer: NodeExec, node_name: str = "local", app: FastAPI = None
+    execer: NodeExec, node_name: str = "local", app: FastAPI = None,
+    token: str = None,
 ) -> FastAPI:
     """Build the agent app, or graft the /agent routes onto an existing app
     (the operator entrypoint serves API + agent surface in one process)."""
+    if token is None:
+        token = os.environ.get("CRO_AGENT_TOKEN", "")
+
+    def authorize(request: Request) -> None:
+        if not token:
+            return
+        auth = request.headers.get("authorization", "")
+        if not hmac.compare_digest(auth, f"Bearer {token}"):
+            raise HTTPException(401, "agent API requires a valid bearer token")
+
     standalone = app is None
     if standalone:
         app = FastAPI(title="cro-amd node agent")
@@ -37,6 +57,7 @@ def build_agent_app(
 
     @app.post("/agent/run")
     async def run(request: Request):
+        authorize(request)
         body = await request.json()
         argv = body.get("argv", [])
         if not argv or not isinstance(argv, list):
@@ -49,7 +70,8 @@ def build_agent_app(
         return {"rc": rc, "stdout": out, "stderr": err}
 
     @app.get("/agent/file")
-    def read_file(path: str):
+    def read_file(path: str, request: Request):
+        authorize(request)
         try:
             return Response(execer.read_file(node_name, path), media_type="text/plain")
         except FileNotFoundError:
@@ -59,6 +81,7 @@ def build_agent_app(
 
     @app.put("/agent/file")
     async def write_file(path: str, request: Request):
+        authorize(request)
         data = (await request.body()).decode()
         try:
             execer.write_file(node_name, path, data)
@@ -67,7 +90,8 @@ def build_agent_app(
         return {"written": path}
 
     @app.get("/agent/dir")
-    def list_dir(path: str):
+    def list_dir(path: str, request: Request):
+        authorize(request)
         try:
             return {"entries": execer.list_dir(node_name, path)}
         except FileNotFoundError:
@@ -76,7 +100,8 @@ def build_agent_app(
             raise HTTPException(403, str(exc))
 
     @app.get("/agent/exists")
-    def exists(path: str):
+    def exists(path: str, request: Request):
+        authorize(request)
         return {"exists": execer.path_exists(node_name, path)}
 
     return app

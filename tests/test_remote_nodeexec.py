@@ -133,3 +133,25 @@ def test_probe_via_exec_failure_shapes(remote_exec):
                         (0, "garbage not json", ""))
     result = probe_via_exec(remote, NODE, gpu)
     assert not result["ok"] and "unparseable" in result["msg"]
+
+
+def test_agent_token_auth():
+    """CRO_AGENT_TOKEN gates every /agent route (the standalone analog of
+    the RBAC around the reference's pods/exec path)."""
+    backend = MockNodeExec()
+    kfd_fixture(backend, 1, node=NODE)
+    app = build_agent_app(backend, node_name=NODE, token="s3cret")
+
+    anon = RemoteNodeExec({NODE: "http://agent"})
+    anon._http = TestClient(app)
+    with pytest.raises(Exception):  # 401 surfaces as HTTPStatusError
+        anon.path_exists(NODE, "/sys/module/amdgpu")
+
+    authed = RemoteNodeExec({NODE: "http://agent"})
+    authed._http = TestClient(app, headers={"Authorization": "Bearer s3cret"})
+    assert authed.path_exists(NODE, "/sys/module/amdgpu")
+
+    wrong = RemoteNodeExec({NODE: "http://agent"})
+    wrong._http = TestClient(app, headers={"Authorization": "Bearer nope"})
+    with pytest.raises(Exception):
+        wrong.read_file(NODE, "/sys/class/kfd/kfd/topology/nodes/1/properties")
