@@ -14,10 +14,10 @@ makes the enum real on the node:
 * drain via PCI hot-remove of the endpoint (the same sysfs mechanism the
   GPU drain uses — CXL memory composition rides the same fabric).
 
-Load checking is conservative-empty: attributing mapped dax pages to
-processes needs /proc/<pid>/smaps scans; composed-memory deployments use
-``force_detach`` or region offlining ahead of detach (documented in
-ROADMAP.md).
+Load checking attributes mapped dax pages to processes by scanning
+``/proc/<pid>/maps`` for the memdev's ``/dev/daxX.Y`` path (the CXL
+analog of the per-GPU KFD vram attribution): a process that mmap'ed the
+dax device holds composed memory and blocks a non-forced detach.
 """
 
 from __future__ import annotations
@@ -143,8 +143,48 @@ class CxlNodeOps:
 
     is_visible_dra = is_visible  # no CXL DRA driver exists yet
 
+    def dax_holders(self, node: str, device_id: str) -> List[int]:
+        """PIDs holding a mapping of the memdev's dax device (via
+        /proc/<pid>/maps — mapped file paths include /dev/daxX.Y)."""
+        dev = self.find(node, device_id)
+        if dev is None or not dev.dax_path:
+            return []
+        holders: List[int] = []
+        try:
+            entries = self.execer.list_dir(node, "/proc")
+        except (FileNotFoundError, PermissionError, OSError):
+            return []
+        for entry in entries:
+            if not entry.isdigit():
+                continue
+            try:
+                maps = self.execer.read_file(node, f"/proc/{entry}/maps")
+            except (FileNotFoundError, PermissionError, OSError):
+                continue  # raced exit / hidden pid — not a holder we can see
+            if dev.dax_path in maps:
+                holders.append(int(entry))
+        return holders
+
     def check_no_loads(self, node: str, device_id: Optional[str] = None) -> None:
-        return None  # see module docstring: dax attribution is future work
+        """Block detach while processes hold composed memory mapped.
+
+        ``device_id=None`` (whole-node, DEVICE_PLUGIN shape) checks every
+        enumerated memdev; per-device otherwise. A memdev with no bound
+        dax region cannot be mmap'ed and passes vacuously.
+        """
+        from .amdgpu import GPULoadsPresent
+
+        targets = (
+            [d.device_id for d in self.enumerate(node)]
+            if device_id is None
+            else [device_id]
+        )
+        for did in targets:
+            holders = self.dax_holders(node, did)
+            if holders:
+                raise GPULoadsPresent(
+                    f"processes hold dax mappings of {did}: {holders}"
+                )
 
     def drain(self, node: str, device_id: str) -> None:
         dev = self.find(node, device_id)

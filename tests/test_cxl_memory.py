@@ -166,3 +166,31 @@ def test_cxlmemory_lifecycle_through_manager(mixed_stack):
     )
     assert mixed_stack.cxl_ops.cdi.devices(NODE) == []
     assert mixed_stack.fabric.attached_to(NODE) == []
+
+
+def test_dax_load_attribution_blocks_detach():
+    """check_no_loads scans /proc/<pid>/maps for the memdev's dax path —
+    the CXL analog of per-GPU KFD vram attribution."""
+    from cro_amd.nodeops.amdgpu import GPULoadsPresent
+
+    ex = MockNodeExec()
+    ids = cxl_fixture(ex, 2)
+    ops = CxlNodeOps(ex)
+    # pid 100 maps mem0's dax; pid 200 maps something else; pid 300 unreadable
+    ex.set_file(NODE, "/proc/100/maps",
+                "7f00-7f10 rw-s 0 00:0e 42 /dev/dax0.0\n")
+    ex.set_file(NODE, "/proc/200/maps",
+                "7f00-7f10 rw-s 0 00:0e 43 /dev/shm/x\n")
+    ex.set_file(NODE, "/proc/self/status", "")  # non-numeric entry ignored
+
+    assert ops.dax_holders(NODE, ids[0]) == [100]
+    assert ops.dax_holders(NODE, ids[1]) == []
+    with pytest.raises(GPULoadsPresent, match=r"\[100\]"):
+        ops.check_no_loads(NODE, ids[0])
+    ops.check_no_loads(NODE, ids[1])  # other memdev unaffected
+    with pytest.raises(GPULoadsPresent):
+        ops.check_no_loads(NODE)  # whole-node form sees mem0's holder
+
+    # holder exits → detach unblocks
+    ex.files.pop((NODE, "/proc/100/maps"))
+    ops.check_no_loads(NODE)
