@@ -125,7 +125,9 @@ class AmdNodeOps(NodeOps):
         # times (refresh/visibility/CDI/probe) and each full KFD sysfs scan
         # costs ~2 ms; lifecycle mutations invalidate explicitly
         self.enum_cache_ttl = 0.025
-        self._enum_cache: Dict[str, tuple] = {}  # node -> (monotonic, gpus)
+        # node -> (monotonic, gpus, generation_id) — TTL bounds staleness,
+        # generation_id revalidation avoids full topology walks after it
+        self._enum_cache: Dict[str, tuple] = {}
         self._enum_lock = threading.Lock()
         # per-device static metadata (VRAM/xGMI/card map) survives the TTL
         # cache — keyed by device identity, see kfd.enumerate_gpus
@@ -144,6 +146,19 @@ class AmdNodeOps(NodeOps):
 
     # -- enumeration / visibility -----------------------------------------
 
+    def _topology_generation(self, node: str) -> Optional[str]:
+        """KFD bumps /sys/class/kfd/kfd/topology/generation_id on every
+        topology change (hot-plug either way) — one tiny sysfs read tells
+        whether a cached enumeration is still exact. None when unreadable
+        (mock fixtures, hidden sysfs) → callers fall back to a full walk."""
+        try:
+            gen = self.execer.read_file(
+                node, "/sys/class/kfd/kfd/topology/generation_id"
+            ).strip()
+            return gen or None
+        except (FileNotFoundError, PermissionError, OSError, ExecError):
+            return None
+
     def enumerate(self, node: str) -> List[GPUDevice]:
         now = time.monotonic()
         with self._enum_lock:
@@ -152,13 +167,21 @@ class AmdNodeOps(NodeOps):
                 gpus = cached[1]
             else:
                 gpus = None
+        if gpus is None and cached is not None and cached[2] is not None:
+            # TTL expired, but an unchanged generation_id proves the
+            # topology did not move — revalidate instead of re-walking
+            if self._topology_generation(node) == cached[2]:
+                gpus = cached[1]
+                with self._enum_lock:
+                    self._enum_cache[node] = (now, gpus, cached[2])
         if gpus is None:
             try:
                 gpus = enumerate_gpus(self.execer, node, self._static_meta)
             except ExecError:
                 gpus = enumerate_gpus_amdsmi(self.execer, node)
+            gen = self._topology_generation(node)
             with self._enum_lock:
-                self._enum_cache[node] = (now, gpus)
+                self._enum_cache[node] = (now, gpus, gen)
         if self.destructive:
             return list(gpus)
         with self._sim_lock:

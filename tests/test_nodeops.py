@@ -457,3 +457,33 @@ def test_static_meta_survives_invalidation():
     # vram banks were NOT re-read (static cache), yet values are present
     assert len(mem_reads_after) == len(mem_reads)
     assert ops.enumerate(NODE)[0].vram_bytes == 309237645312
+
+
+def test_enum_generation_id_revalidation():
+    """After the TTL, an unchanged /sys/class/kfd/kfd/topology/generation_id
+    revalidates the cached enumeration (one sysfs read, no topology walk);
+    a bumped generation_id forces the full walk."""
+    ex = MockNodeExec()
+    ids = kfd_fixture(ex, 2)
+    gen_path = "/sys/class/kfd/kfd/topology/generation_id"
+    ex.set_file(NODE, gen_path, "7\n")
+    ops = AmdNodeOps(ex, cdi_dir="/etc/cdi", destructive=True)
+    ops.enum_cache_ttl = 0.0  # every call is past the TTL
+
+    assert [g.device_id for g in ops.enumerate(NODE)] == ids
+    # break the topology files; unchanged generation_id → cache still served
+    saved = {k: v for k, v in ex.files.items() if "topology/nodes" in k[1]}
+    for k in saved:
+        ex.files.pop(k)
+    assert [g.device_id for g in ops.enumerate(NODE)] == ids
+
+    # rebuild with one fewer GPU and bump the generation → full re-walk sees it
+    kfd_fixture(ex, 1)
+    ex.set_file(NODE, gen_path, "8\n")
+    # clear per-device static caches tied to the old tree
+    ops._static_meta.clear()
+    assert len(ops.enumerate(NODE)) == 1
+
+    # explicit invalidation always forces a walk even with stable gen
+    ops._invalidate_enum(NODE)
+    assert len(ops.enumerate(NODE)) == 1
