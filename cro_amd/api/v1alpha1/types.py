@@ -242,6 +242,27 @@ class DaemonSet(K8sObject):
     status: DaemonSetStatus = Field(default_factory=DaemonSetStatus)
 
 
+class DeviceConfigDriver(_Model):
+    enable: bool = False
+    daemonset_name: str = "amd-gpu-driver"
+
+
+class DeviceConfigSpec(_Model):
+    driver: DeviceConfigDriver = Field(default_factory=DeviceConfigDriver)
+
+
+class DeviceConfig(K8sObject):
+    """Minimal AMD GPU operator DeviceConfig surface — the ClusterPolicy
+    analog in the reference's driver-detection chain (gpus.go:97-127:
+    driver daemonset pod → Container; ClusterPolicy driver enabled →
+    Container; else modinfo → Host). Only the containerized-driver gate is
+    modeled; the full CRD belongs to the AMD GPU operator."""
+
+    KIND: ClassVar[str] = "DeviceConfig"
+    apiVersion: str = "amd.com/v1alpha1"
+    spec: DeviceConfigSpec = Field(default_factory=DeviceConfigSpec)
+
+
 class Event(K8sObject):
     """Lifecycle event stream (beyond the reference, which emits no
     Kubernetes Events anywhere — no EventRecorder in any controller).
@@ -272,6 +293,7 @@ ALL_KINDS = {
         ResourceSlice,
         DeviceTaintRule,
         DaemonSet,
+        DeviceConfig,
         Event,
     )
 }

@@ -135,14 +135,64 @@ class AmdNodeOps(NodeOps):
 
     # -- driver ------------------------------------------------------------
 
-    def ensure_driver(self, node: str) -> None:
-        """Host-driver detection: /sys/module/amdgpu (gpus.go:97-127 analog).
+    def driver_mode(self, node: str) -> str:
+        """Driver-detection chain (gpus.go:97-127 analog): an AMD GPU
+        operator DeviceConfig with the containerized driver enabled →
+        ``container`` (the ClusterPolicy arm); else /sys/module/amdgpu on
+        the node → ``host`` (the modinfo arm); else ``none``."""
+        if self.client is not None:
+            from ..api.v1alpha1.types import DeviceConfig
 
-        A driver running in a container would surface through a different
-        sysroot on the execer; either way the module directory must exist.
+            try:
+                configs = self.client.list(DeviceConfig)
+            except Exception:
+                configs = []
+            if any(dc.spec.driver.enable for dc in configs):
+                return "container"
+        if self.execer.path_exists(node, "/sys/module/amdgpu"):
+            return "host"
+        return "none"
+
+    def ensure_driver(self, node: str) -> None:
+        """Driver gate before any attach work.
+
+        Host mode: the amdgpu module directory must exist. Container mode:
+        the driver daemonset must be fully ready (the driver-pod readiness
+        gate, gpus.go:161-193) AND the module loaded — a containerized
+        driver still surfaces /sys/module/amdgpu on the host.
         """
-        if not self.execer.path_exists(node, "/sys/module/amdgpu"):
+        mode = self.driver_mode(node)
+        if mode == "none":
             raise DriverMissing(f"amdgpu kernel module not loaded on node {node}")
+        if mode == "container":
+            import os
+
+            from ..api.v1alpha1.types import DaemonSet, DeviceConfig
+            from ..runtime.errors import NotFoundError
+
+            ns = os.environ.get("CRO_AMD_GPU_OPERATOR_NAMESPACE", "amd-gpu-operator")
+            ds_name = "amd-gpu-driver"
+            for dc in self.client.list(DeviceConfig):
+                if dc.spec.driver.enable and dc.spec.driver.daemonset_name:
+                    ds_name = dc.spec.driver.daemonset_name
+                    break
+            try:
+                ds = self.client.get(DaemonSet, f"{ns}/{ds_name}")
+            except NotFoundError:
+                raise DriverMissing(
+                    f"containerized driver enabled but daemonset "
+                    f"{ns}/{ds_name} not found"
+                )
+            st = ds.status
+            if st.number_ready < st.desired_number_scheduled:
+                raise DriverMissing(
+                    f"driver daemonset {ns}/{ds_name} not ready "
+                    f"({st.number_ready}/{st.desired_number_scheduled})"
+                )
+            if not self.execer.path_exists(node, "/sys/module/amdgpu"):
+                raise DriverMissing(
+                    f"driver daemonset ready but amdgpu not loaded on {node}"
+                )
 
     # -- enumeration / visibility -----------------------------------------
 

@@ -487,3 +487,55 @@ def test_enum_generation_id_revalidation():
     # explicit invalidation always forces a walk even with stable gen
     ops._invalidate_enum(NODE)
     assert len(ops.enumerate(NODE)) == 1
+
+
+def test_driver_detection_chain():
+    """DeviceConfig (container driver) → daemonset readiness gate;
+    no DeviceConfig → host /sys/module/amdgpu; neither → DriverMissing
+    (gpus.go:97-193 chain)."""
+    from cro_amd.api.v1alpha1.types import DaemonSet, DeviceConfig
+    from cro_amd.runtime.client import Client
+    from cro_amd.runtime.store import InMemoryStore
+
+    client = Client(InMemoryStore())
+    ex = MockNodeExec()
+    kfd_fixture(ex, 1)
+    ops = AmdNodeOps(ex, client=client, cdi_dir="/etc/cdi")
+
+    # host mode: module present, no DeviceConfig
+    assert ops.driver_mode(NODE) == "host"
+    ops.ensure_driver(NODE)
+
+    # container mode: DeviceConfig enables the driver → daemonset must exist
+    dc = DeviceConfig()
+    dc.metadata.name = "default"
+    dc.spec.driver.enable = True
+    client.create(dc)
+    assert ops.driver_mode(NODE) == "container"
+    with pytest.raises(DriverMissing, match="not found"):
+        ops.ensure_driver(NODE)
+
+    ds = DaemonSet()
+    ds.metadata.name = "amd-gpu-operator/amd-gpu-driver"
+    ds.status.desired_number_scheduled = 1
+    ds.status.number_ready = 0
+    client.create(ds)
+    with pytest.raises(DriverMissing, match="not ready"):
+        ops.ensure_driver(NODE)
+
+    ds = client.get(DaemonSet, "amd-gpu-operator/amd-gpu-driver")
+    ds.status.number_ready = 1
+    client.update_status(ds)
+    ops.ensure_driver(NODE)  # ready + module loaded → passes
+
+    # module gone ⇒ even a ready containerized driver fails the gate
+    ex.files.pop((NODE, "/sys/module/amdgpu/version"), None)
+    for k in [k for k in list(ex.dirs) if "module/amdgpu" in k[1]] if hasattr(ex, "dirs") else []:
+        ex.dirs.pop(k)
+    if not ex.path_exists(NODE, "/sys/module/amdgpu"):
+        with pytest.raises(DriverMissing, match="not loaded"):
+            ops.ensure_driver(NODE)
+
+    # none: fresh ops without module or DeviceConfig
+    bare = AmdNodeOps(MockNodeExec(), cdi_dir="/etc/cdi")
+    assert bare.driver_mode(NODE) == "none"
