@@ -68,9 +68,37 @@ def _raise_for(resp: httpx.Response) -> None:
 
 
 class RemoteClient:
-    def __init__(self, base_url: str, transport: Optional[httpx.BaseTransport] = None):
+    """HTTP Client over the apiserver-shaped REST API.
+
+    The URL scheme is the kube-apiserver scheme for cluster-scoped CRDs
+    (``/apis/cro.hpsys.ibm.ie.com/v1alpha1/<plural>[/<name>[/status]]``),
+    so the same client reaches a real kube-apiserver given credentials:
+    ``token`` (or ``CRO_API_TOKEN``) sends ``Authorization: Bearer`` —
+    the in-cluster ServiceAccount token — and ``ca`` (or ``CRO_API_CA``)
+    verifies the serving cert (``verify=False`` to skip)."""
+
+    def __init__(
+        self,
+        base_url: str,
+        transport: Optional[httpx.BaseTransport] = None,
+        token: Optional[str] = None,
+        ca: Optional[str] = None,
+        verify: Union[bool, str, None] = None,
+    ):
+        import os
+
         self.base_url = base_url.rstrip("/")
-        self._http = httpx.Client(base_url=self.base_url, transport=transport, timeout=30)
+        if token is None:
+            token = os.environ.get("CRO_API_TOKEN", "")
+        if ca is None:
+            ca = os.environ.get("CRO_API_CA", "")
+        if verify is None:
+            verify = ca if ca else True
+        headers = {"Authorization": f"Bearer {token}"} if token else None
+        self._http = httpx.Client(
+            base_url=self.base_url, transport=transport, timeout=30,
+            headers=headers, verify=verify,
+        )
         self._stop = threading.Event()
         self._watch_threads: List[threading.Thread] = []
 

@@ -251,3 +251,20 @@ def test_croctl_watch_over_http(api_server, capsys):
     assert len(lines) == 2
     assert all(ln.startswith("ADDED") for ln in lines)
     assert {"w1", "w2"} <= {ln.split()[1] for ln in lines}
+
+
+def test_remote_client_credentials(monkeypatch):
+    """Bearer token + CA plumbing for real kube-apiserver use (the URL
+    scheme already matches cluster-scoped CRD paths)."""
+    from cro_amd.runtime.remote import RemoteClient
+
+    c = RemoteClient("http://api", token="sa-token")
+    assert c._http.headers["authorization"] == "Bearer sa-token"
+
+    monkeypatch.setenv("CRO_API_TOKEN", "env-token")
+    c2 = RemoteClient("http://api")
+    assert c2._http.headers["authorization"] == "Bearer env-token"
+
+    monkeypatch.delenv("CRO_API_TOKEN")
+    c3 = RemoteClient("http://api")
+    assert "authorization" not in c3._http.headers
