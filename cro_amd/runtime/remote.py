@@ -182,12 +182,20 @@ class RemoteClient:
 
     def _watch_kind(self, kind: str, q: "queue.Queue[WatchEvent]") -> None:
         cls, plural = self._resolve(kind)
-        # the stream's initial list doubles as the informer-cache replay;
-        # reconnects re-list (level-triggered reconciles tolerate replays)
+        # the stream's initial list doubles as the informer-cache replay.
+        # Every event line carries an ``rv`` resume token; reconnects pass
+        # the last one back so the server replays only missed events from
+        # its watch-cache — an aged-out token gets an in-stream
+        # ERROR/Expired followed by a full re-list (level-triggered
+        # reconciles tolerate the replay either way).
+        last_rv = -1
         while not self._stop.is_set():
             try:
+                params = {"watch": "true"}
+                if last_rv >= 0:
+                    params["resourceVersion"] = str(last_rv)
                 with self._http.stream(
-                    "GET", f"{BASE}/{plural}", params={"watch": "true"}, timeout=None
+                    "GET", f"{BASE}/{plural}", params=params, timeout=None
                 ) as resp:
                     for line in resp.iter_lines():
                         if self._stop.is_set():
@@ -195,6 +203,10 @@ class RemoteClient:
                         if not line.strip():
                             continue  # keepalive
                         ev = json.loads(line)
+                        if ev["type"] == "ERROR":
+                            last_rv = -1  # token expired; full replay follows
+                            continue
+                        last_rv = int(ev.get("rv", last_rv))
                         q.put(WatchEvent(ev["type"], cls.model_validate(ev["object"])))
             except Exception as exc:
                 if self._stop.is_set():

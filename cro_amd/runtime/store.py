@@ -33,6 +33,7 @@ import queue
 import threading
 import time
 import uuid as uuidlib
+from collections import deque
 from dataclasses import dataclass
 from typing import Callable, Dict, List, Optional
 
@@ -50,6 +51,7 @@ class WatchEvent:
     type: str  # ADDED | MODIFIED | DELETED
     object: K8sObject
     old_object: Optional[K8sObject] = None
+    seq: int = 0  # global event sequence — the watch resume token
 
 
 # admission validator: fn(operation: "CREATE"|"UPDATE", old: obj|None, new: obj)
@@ -79,6 +81,11 @@ class InMemoryStore:
         self._rv = itertools.count(1)
         self._watchers: List[_Watcher] = []
         self._admission: Dict[str, List[AdmissionFn]] = {}
+        # bounded event history for resourceVersion-resuming watches
+        # (apiserver watch-cache analog): a reconnecting client replays
+        # only what it missed instead of a full re-list
+        self._event_seq = 0
+        self._event_log: "deque[WatchEvent]" = deque(maxlen=4096)
 
     # -- admission ---------------------------------------------------------
 
@@ -106,7 +113,36 @@ class InMemoryStore:
                     w.closed = True
             self._watchers = [w for w in self._watchers if not w.closed]
 
+    def current_seq(self) -> int:
+        with self._lock:
+            return self._event_seq
+
+    def events_since(
+        self, seq: int, kinds: Optional[List[str]] = None
+    ) -> Optional[List[WatchEvent]]:
+        """Events after resume token ``seq``, or None when the token has
+        aged out of the bounded log (client must fall back to re-list)."""
+        wanted = set(kinds) if kinds else None
+        with self._lock:
+            if seq >= self._event_seq:
+                return []
+            if not self._event_log or self._event_log[0].seq > seq + 1:
+                return None  # compacted past the token
+            return [
+                WatchEvent(e.type, e.object.model_copy(deep=True), None, e.seq)
+                for e in self._event_log
+                if e.seq > seq and (wanted is None or e.object.kind in wanted)
+            ]
+
     def _notify(self, ev: WatchEvent) -> None:
+        # callers hold self._lock (all mutators notify inside their
+        # critical section), so the seq assignment and log append are
+        # atomic with the mutation itself
+        self._event_seq += 1
+        ev.seq = self._event_seq
+        self._event_log.append(
+            WatchEvent(ev.type, ev.object.model_copy(deep=True), None, ev.seq)
+        )
         # deep copies per subscriber so no watcher can mutate shared state
         for w in self._watchers:
             if not w.closed and w.wants(ev.object.kind):
@@ -115,6 +151,7 @@ class InMemoryStore:
                         ev.type,
                         ev.object.model_copy(deep=True),
                         ev.old_object.model_copy(deep=True) if ev.old_object else None,
+                        ev.seq,
                     )
                 )
 

@@ -107,10 +107,15 @@ def build_app(client: Client) -> FastAPI:
         )
 
     @app.get(BASE + "/{plural}")
-    async def list_objects(plural: str, labelSelector: str = "", watch: bool = False):
+    async def list_objects(
+        plural: str,
+        labelSelector: str = "",
+        watch: bool = False,
+        resourceVersion: int = -1,
+    ):
         cls = _cls(plural)
         if watch:
-            return await _watch_stream(cls)
+            return await _watch_stream(cls, resourceVersion)
         labels = None
         if labelSelector:
             labels = dict(part.split("=", 1) for part in labelSelector.split(","))
@@ -121,9 +126,13 @@ def build_app(client: Client) -> FastAPI:
             "items": [_dump(o) for o in items],
         }
 
-    async def _watch_stream(cls):
-        """k8s-style list+watch: current objects as synthetic ADDED events,
-        then live events, newline-delimited JSON."""
+    async def _watch_stream(cls, resource_version: int = -1):
+        """k8s-style list+watch, newline-delimited JSON. Every event line
+        carries ``rv`` (the store's event sequence) as a resume token:
+        reconnecting with ``?resourceVersion=<rv>`` replays only missed
+        events from the bounded watch-cache; an aged-out token gets one
+        ``{"type": "ERROR", "reason": "Expired"}`` line followed by the
+        full list replay (the apiserver's 410-Gone contract, in-stream)."""
         import asyncio
         import json as _json
         import queue as _queue
@@ -132,9 +141,26 @@ def build_app(client: Client) -> FastAPI:
 
         async def gen():
             events = client.watch([cls.KIND])
+            store = getattr(client, "store", None)
+            last_seq = 0
             try:
-                for obj in client.list(cls):
-                    yield _json.dumps({"type": "ADDED", "object": _dump(obj)}) + "\n"
+                buffered = None
+                if resource_version >= 0 and store is not None:
+                    buffered = store.events_since(resource_version, [cls.KIND])
+                if buffered is not None:
+                    for ev in buffered:
+                        last_seq = ev.seq
+                        yield _json.dumps(
+                            {"type": ev.type, "object": _dump(ev.object), "rv": ev.seq}
+                        ) + "\n"
+                else:
+                    if resource_version >= 0:
+                        yield _json.dumps({"type": "ERROR", "reason": "Expired"}) + "\n"
+                    snapshot_rv = store.current_seq() if store is not None else 0
+                    for obj in client.list(cls):
+                        yield _json.dumps(
+                            {"type": "ADDED", "object": _dump(obj), "rv": snapshot_rv}
+                        ) + "\n"
                 loop = asyncio.get_running_loop()
                 while True:
                     try:
@@ -144,7 +170,12 @@ def build_app(client: Client) -> FastAPI:
                         continue
                     except RuntimeError:
                         return  # event loop / executor shutting down
-                    yield _json.dumps({"type": ev.type, "object": _dump(ev.object)}) + "\n"
+                    if ev.seq and ev.seq <= last_seq:
+                        continue  # already served from the resume buffer
+                    last_seq = ev.seq or last_seq
+                    yield _json.dumps(
+                        {"type": ev.type, "object": _dump(ev.object), "rv": ev.seq}
+                    ) + "\n"
             finally:
                 # disconnects must release the watcher or every later event
                 # fans out to dead queues forever

@@ -268,3 +268,77 @@ def test_remote_client_credentials(monkeypatch):
     monkeypatch.delenv("CRO_API_TOKEN")
     c3 = RemoteClient("http://api")
     assert "authorization" not in c3._http.headers
+
+
+def test_watch_resume_token(api_server):
+    """Reconnecting with ?resourceVersion=<rv> replays only missed events
+    from the server watch-cache; an aged-out token gets ERROR/Expired then
+    a full list replay (in-stream 410-Gone contract)."""
+    import json
+
+    import httpx
+
+    url, server_mgr = api_server
+    server_mgr.client.create(make_request("a", target_node="node0"))
+    server_mgr.client.create(make_request("b", model="mi300x", target_node="node0"))
+
+    base = f"{url}/apis/cro.hpsys.ibm.ie.com/v1alpha1/composabilityrequests"
+    with httpx.stream("GET", base, params={"watch": "true"}, timeout=10) as resp:
+        lines = []
+        for line in resp.iter_lines():
+            if line.strip():
+                lines.append(json.loads(line))
+            if len(lines) == 2:
+                break
+    assert [l["object"]["metadata"]["name"] for l in lines] == ["a", "b"]
+    rv = lines[-1]["rv"]
+
+    # miss one event while disconnected
+    server_mgr.client.create(
+        make_request("c", model="mi308x", target_node="node0")
+    )
+    with httpx.stream(
+        "GET", base, params={"watch": "true", "resourceVersion": str(rv)}, timeout=10
+    ) as resp:
+        first = None
+        for line in resp.iter_lines():
+            if line.strip():
+                first = json.loads(line)
+                break
+    assert first["type"] == "ADDED"
+    assert first["object"]["metadata"]["name"] == "c"  # ONLY the missed event
+    assert first["rv"] > rv
+
+    # aged-out token → ERROR then full replay
+    store = server_mgr.store
+    while store._event_log:
+        store._event_log.popleft()
+    with httpx.stream(
+        "GET", base, params={"watch": "true", "resourceVersion": "1"}, timeout=10
+    ) as resp:
+        got = []
+        for line in resp.iter_lines():
+            if line.strip():
+                got.append(json.loads(line))
+            if len(got) == 4:
+                break
+    assert got[0] == {"type": "ERROR", "reason": "Expired"}
+    assert sorted(l["object"]["metadata"]["name"] for l in got[1:]) == ["a", "b", "c"]
+
+
+def test_store_events_since_semantics():
+    from cro_amd.runtime.store import InMemoryStore
+    from cro_amd.runtime.client import Client
+
+    store = InMemoryStore()
+    client = Client(store)
+    assert store.events_since(0) == []  # nothing happened yet
+    client.create(make_request("x", target_node="n"))
+    evs = store.events_since(0)
+    assert len(evs) == 1 and evs[0].object.metadata.name == "x"
+    assert store.events_since(evs[0].seq) == []
+    assert store.events_since(0, kinds=["ComposableResource"]) == []
+    # compaction gap → None
+    store._event_log.popleft()
+    client.create(make_request("y", model="m2", target_node="n"))
+    assert store.events_since(0) is None
