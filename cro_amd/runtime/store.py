@@ -124,8 +124,12 @@ class InMemoryStore:
         aged out of the bounded log (client must fall back to re-list)."""
         wanted = set(kinds) if kinds else None
         with self._lock:
-            if seq >= self._event_seq:
-                return []
+            if seq == self._event_seq:
+                return []  # caught up
+            if seq > self._event_seq:
+                # token from a previous store incarnation (server restarted
+                # with fresh state) — not comparable; force a full re-list
+                return None
             if not self._event_log or self._event_log[0].seq > seq + 1:
                 return None  # compacted past the token
             return [

@@ -342,3 +342,16 @@ def test_store_events_since_semantics():
     store._event_log.popleft()
     client.create(make_request("y", model="m2", target_node="n"))
     assert store.events_since(0) is None
+
+
+def test_events_since_foreign_token_forces_relist():
+    """A resume token larger than the store's own sequence (server
+    restarted with fresh state) is not comparable — must expire, not
+    silently return 'caught up'."""
+    from cro_amd.runtime.client import Client
+    from cro_amd.runtime.store import InMemoryStore
+
+    store = InMemoryStore()
+    Client(store).create(make_request("x", target_node="n"))
+    assert store.events_since(999) is None
+    assert store.events_since(store.current_seq()) == []
