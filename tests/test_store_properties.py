@@ -163,6 +163,26 @@ class StoreMachine(RuleBasedStateMachine):
         stored = {o.metadata.name for o in self.store.list("ComposableResource")}
         assert stored == set(self.model), (stored, set(self.model))
 
+    @invariant()
+    def event_log_replays_to_current_state(self):
+        """Applying events_since(token) over any token must reconstruct
+        the live object set — the correctness condition for resumable
+        watches. Token 0 is valid while the log hasn't compacted."""
+        events = self.store.events_since(0, ["ComposableResource"])
+        if events is None:
+            return  # compacted; resume would fall back to re-list
+        alive = {}
+        for ev in events:
+            if ev.type == "DELETED":
+                alive.pop(ev.object.metadata.name, None)
+            else:
+                alive[ev.object.metadata.name] = ev.object
+        stored = {o.metadata.name for o in self.store.list("ComposableResource")}
+        assert set(alive) == stored, (set(alive), stored)
+        # seqs strictly increase
+        seqs = [ev.seq for ev in events]
+        assert seqs == sorted(seqs) and len(set(seqs)) == len(seqs)
+
 
 StoreMachine.TestCase.settings = settings(max_examples=30, stateful_step_count=40, deadline=None)
 TestStoreProperties = StoreMachine.TestCase
