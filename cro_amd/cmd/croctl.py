@@ -212,21 +212,35 @@ def main(argv=None, client: httpx.Client = None) -> int:
         return resp
 
     if args.command == "apply":
-        with open(args.filename) as f:
-            obj = yaml.safe_load(f)
-        plural = obj["kind"].lower() + "s"
-        name = obj["metadata"]["name"]
-        if http.get(f"{BASE}/{plural}/{name}").status_code == 200:
-            resp = update_with_conflict_retry(
-                plural, name, lambda cur: cur.update({"spec": obj["spec"]})
+        # kubectl shape: -f FILE (multi-document YAML) or -f DIRECTORY
+        paths = [args.filename]
+        if os.path.isdir(args.filename):
+            paths = sorted(
+                os.path.join(args.filename, p)
+                for p in os.listdir(args.filename)
+                if p.endswith((".yaml", ".yml", ".json"))
             )
-            verb = "configured"
-        else:
-            resp = http.post(f"{BASE}/{plural}", json=obj)
-            verb = "created"
-        if resp.status_code not in (200, 201):
-            return fail(resp)
-        print(f"{plural}/{name} {verb}")
+        docs = []
+        for path in paths:
+            with open(path) as f:
+                docs.extend(d for d in yaml.safe_load_all(f) if d)
+        if not docs:
+            print("error: no objects to apply", file=sys.stderr)
+            return 1
+        for obj in docs:
+            plural = obj["kind"].lower() + "s"
+            name = obj["metadata"]["name"]
+            if http.get(f"{BASE}/{plural}/{name}").status_code == 200:
+                resp = update_with_conflict_retry(
+                    plural, name, lambda cur, o=obj: cur.update({"spec": o["spec"]})
+                )
+                verb = "configured"
+            else:
+                resp = http.post(f"{BASE}/{plural}", json=obj)
+                verb = "created"
+            if resp.status_code not in (200, 201):
+                return fail(resp)
+            print(f"{plural}/{name} {verb}")
         return 0
 
     if args.command == "delete":

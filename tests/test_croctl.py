@@ -114,3 +114,42 @@ def test_events_after_lifecycle(cli, tmp_path):
     assert rc == 0 and "NodesAllocated" in out and "Online" in out
     rc, out, _ = run("events", "--for", "ComposabilityRequest/e1")
     assert rc == 0 and "Running" in out
+
+
+def test_apply_multidoc_and_directory(cli, tmp_path):
+    """kubectl parity: -f with multi-document YAML and with a directory."""
+    multi = tmp_path / "pair.yaml"
+    multi.write_text(
+        "\n---\n".join(
+            yaml.safe_dump(
+                {
+                    "apiVersion": "cro.hpsys.ibm.ie.com/v1alpha1",
+                    "kind": "ComposabilityRequest",
+                    "metadata": {"name": n},
+                    "spec": {"resource": {
+                        "type": "gpu", "model": m, "size": 1,
+                        "target_node": "node0",
+                    }},
+                }
+            )
+            for n, m in (("md1", "mi355x"), ("md2", "mi300x"))
+        )
+    )
+    run, stack = cli
+    rc, out, _ = run("apply", "-f", str(multi))
+    assert rc == 0
+    assert "md1 created" in out and "md2 created" in out
+
+    d = tmp_path / "bundle"
+    d.mkdir()
+    (d / "one.yaml").write_text(
+        yaml.safe_dump({
+            "apiVersion": "cro.hpsys.ibm.ie.com/v1alpha1",
+            "kind": "ComposabilityRequest",
+            "metadata": {"name": "md3"},
+            "spec": {"resource": {"type": "gpu", "model": "mi308x",
+                                  "size": 1, "target_node": "node0"}},
+        })
+    )
+    rc, out, _ = run("apply", "-f", str(d))
+    assert rc == 0 and "md3 created" in out
