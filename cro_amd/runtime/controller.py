@@ -81,6 +81,8 @@ class Controller:
         controllers are watching, or cross-controller events are lost."""
         kinds = sorted({s.kind for s in self.sources})
         events = store.watch(kinds)
+        self._store = store
+        self._events = events
 
         # replay (controller-runtime parity): every object already in the
         # store is delivered as a synthetic ADDED event so a restarted
@@ -157,3 +159,8 @@ class Controller:
         self.queue.shutdown()
         for t in self._threads:
             t.join(timeout=2)
+        # unsubscribe the watch queue: a store outliving this controller
+        # (stack rebuilt on the same store) must not fan out to dead queues
+        store = getattr(self, "_store", None)
+        if store is not None and hasattr(store, "stop_watch"):
+            store.stop_watch(self._events)
