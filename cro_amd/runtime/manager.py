@@ -100,6 +100,9 @@ class Manager:
             c.stop()
         for t in self._threads:
             t.join(timeout=2)
+        recorder = getattr(self, "recorder", None)
+        if recorder is not None:  # drain buffered events before exit
+            recorder.flush(timeout=2.0)
 
     # -- helpers for tests/benches ----------------------------------------
 
