@@ -49,6 +49,15 @@ def validate_composability_request(
                     f"{spec.type} and model {spec.model} already exists"
                 )
     elif spec.allocation_policy == "samenode":
+        # resolve the INCOMING request's implicit target from its status
+        # too (webhook :107-128 resolves both sides): an UPDATE of an
+        # already-allocated no-target request must collide with an explicit
+        # request on the node it actually occupies
+        my_target = spec.target_node
+        if not my_target:
+            for v in request.status.resources.values():
+                my_target = v.node_name
+                break
         for other in existing:
             if other.metadata.name == request.metadata.name or other.spec is None:
                 continue
@@ -58,7 +67,7 @@ def validate_composability_request(
                 for v in other.status.resources.values():
                     target = v.node_name
                     break
-            if target == spec.target_node and o.type == spec.type and o.model == spec.model:
+            if target == my_target and o.type == spec.type and o.model == spec.model:
                 return (
                     f"composabilityRequest resource {other.metadata.name} with type "
                     f"{spec.type} and model {spec.model} already exists"

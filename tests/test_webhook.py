@@ -7,7 +7,7 @@ import pytest
 
 from cro_amd.api.v1alpha1.types import ComposabilityRequest, ScalarResourceStatus
 from cro_amd.runtime.errors import AdmissionDenied
-from cro_amd.webhook.validator import admission_validator
+from cro_amd.webhook.validator import admission_validator, validate_composability_request
 from tests.conftest import make_request
 
 
@@ -77,3 +77,19 @@ def test_two_unpinned_samenode_requests_conflict(guarded_client):
         guarded_client.create(make_request("r2"))
     # a different model is fine
     guarded_client.create(make_request("r3", model="mi300x"))
+
+
+def test_samenode_update_resolves_incoming_implicit_target(client):
+    """An UPDATE of an allocated no-target samenode request collides with
+    an explicit request for the node it actually occupies (both sides of
+    rule 3 resolve implicit targets, webhook :107-128)."""
+    mine = make_request("mine", size=1)  # no target_node
+    mine.status.resources["gpu-x"] = ScalarResourceStatus(node_name="node7")
+    explicit = make_request("explicit", size=1, target_node="node7")
+
+    msg = validate_composability_request(mine, [explicit])
+    assert msg is not None and "explicit" in msg
+
+    # different node → no collision
+    elsewhere = make_request("elsewhere", size=1, target_node="node8")
+    assert validate_composability_request(mine, [elsewhere]) is None
