@@ -154,6 +154,15 @@ class MockFabric(FabricProvider):
 
     def get_resources(self) -> List[DeviceInfo]:
         with self._lock:
+            now = time.monotonic()
+            for d in self._pool.values():
+                # an async compose completes fabric-side whether or not the
+                # requester is still around (CM resize semantics): once the
+                # deadline passes the device is attached ground truth — the
+                # syncer must see it or an abandoned mid-compose attach
+                # (CR deleted while WaitingDeviceAttaching) leaks forever
+                if d.attached_node and d.attach_deadline and now >= d.attach_deadline:
+                    d.attach_deadline = 0.0
             return [
                 DeviceInfo(
                     node_name=d.attached_node,

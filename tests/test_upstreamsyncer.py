@@ -98,3 +98,52 @@ def test_detach_cr_drives_full_physical_detach(mock_world):
     rec.reconcile(name)  # finalizer off → gone
     assert mock_world.client.list(ComposableResource) == []
     assert mock_world.fabric.attached_to("node0") == []
+
+
+def test_abandoned_async_compose_repaired(mock_world):
+    """A CR deleted while its async compose is in flight leaks the device
+    on the fabric (nobody records the identity); once the compose lands,
+    get_resources must surface it so the syncer repairs the leak."""
+    import time
+
+    from cro_amd.api.v1alpha1.types import ComposabilityRequest
+    from cro_amd.fabric.mock import MockFabricConfig
+    from tests.conftest import drive, make_request
+
+    w = mock_world
+    make_node(w.client, "node0")
+    w.fabric.config.asynchronous = True
+    w.fabric.config.attach_latency = 0.05
+    w.ops.set_driver("node0", True)
+
+    w.client.create(make_request("r1", size=1, target_node="node0"))
+    drive(w.request_rec, "r1", n=3)
+    req = w.client.get(ComposabilityRequest, "r1")
+    child = next(iter(req.status.resources))
+    w.resource_rec.reconcile(child)  # Attaching entry
+    w.resource_rec.reconcile(child)  # starts the compose → Waiting
+    assert w.fabric.attached_to("node0")  # fabric holds it mid-compose
+
+    # abandon: delete the request, drain everything before the compose lands
+    w.client.delete(ComposabilityRequest, "r1")
+    for _ in range(10):
+        drive(w.request_rec, "r1", n=2)
+        drive(w.resource_rec, child, n=2)
+        drive(w.request_rec, child, n=2)
+    assert w.client.list(ComposableResource) == []
+    assert w.fabric.attached_to("node0")  # leaked
+
+    time.sleep(0.06)  # compose lands fabric-side
+    # on hardware the composed device is now physically enumerable on the
+    # node; the mock node-ops needs that told explicitly
+    w.ops.fabric_composed("node0", w.fabric.attached_to("node0")[0])
+    syncer = make_syncer(w, grace=0.0)
+    syncer.sync()
+    time.sleep(0.01)
+    syncer.sync()  # grace exceeded → detach CR
+    crs = w.client.list(ComposableResource)
+    assert len(crs) == 1
+    for _ in range(10):
+        drive(w.resource_rec, crs[0].metadata.name, n=3)
+        drive(w.request_rec, crs[0].metadata.name, n=2)
+    assert w.fabric.attached_to("node0") == []  # repaired
