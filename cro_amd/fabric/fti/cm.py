@@ -67,6 +67,11 @@ class FTICMClient(FabricProvider):
         self.transport = transport
         self.verify = verify
         self.token = token or CachedToken(endpoint, transport=transport, verify=verify)
+        # persistent connection pool: health checks run every 30 s per
+        # device — a TLS handshake per call would dominate fabric RTT
+        self._http = httpx.Client(
+            transport=transport, verify=verify, timeout=CM_REQUEST_TIMEOUT
+        )
 
     # -- HTTP plumbing -----------------------------------------------------
 
@@ -82,10 +87,7 @@ class FTICMClient(FabricProvider):
             "Authorization": f"Bearer {self.token.get_token()}",
             "Content-Type": "application/json",
         }
-        with httpx.Client(
-            transport=self.transport, verify=self.verify, timeout=CM_REQUEST_TIMEOUT
-        ) as client:
-            return client.request(method, url, json=json_body, headers=headers)
+        return self._http.request(method, url, json=json_body, headers=headers)
 
     def _get_machine_info(self, machine_id: str) -> dict:
         resp = self._request("GET", self._machine_path(machine_id))

@@ -61,6 +61,10 @@ class FTIFMClient(FabricProvider):
         self.transport = transport
         self.verify = verify
         self.token = token or CachedToken(endpoint, transport=transport, verify=verify)
+        # persistent connection pool (see cm.py)
+        self._http = httpx.Client(
+            transport=transport, verify=verify, timeout=FM_REQUEST_TIMEOUT
+        )
 
     # -- HTTP plumbing -----------------------------------------------------
 
@@ -70,13 +74,10 @@ class FTIFMClient(FabricProvider):
             "Authorization": f"Bearer {self.token.get_token()}",
             "Content-Type": "application/json",
         }
-        with httpx.Client(
-            transport=self.transport, verify=self.verify, timeout=FM_REQUEST_TIMEOUT
-        ) as client:
-            return client.request(
-                method, url, params={"tenant_uuid": self.tenant_id},
-                json=json_body, headers=headers,
-            )
+        return self._http.request(
+            method, url, params={"tenant_uuid": self.tenant_id},
+            json=json_body, headers=headers,
+        )
 
     def _machine_id(self, node_name: str) -> str:
         return resolve_machine_id(self.client, node_name, self.cluster_id)

@@ -110,12 +110,12 @@ class NECClient(FabricProvider):
             else float(os.environ.get("CRO_NEC_POLL_INTERVAL", DEFAULT_POLL_INTERVAL))
         )
         self.poll_attempts = poll_attempts
+        self._http = httpx.Client(transport=transport, timeout=REQUEST_TIMEOUT)
 
     # -- HTTP --------------------------------------------------------------
 
     def _do(self, endpoint: str, method: str, path: str, payload=None) -> dict:
-        with httpx.Client(transport=self.transport, timeout=REQUEST_TIMEOUT) as client:
-            resp = client.request(method, endpoint + path, json=payload)
+        resp = self._http.request(method, endpoint + path, json=payload)
         if not 200 <= resp.status_code < 300:
             raise FabricError(
                 f"request failed: method={method} path={path} "

@@ -43,11 +43,11 @@ class SunfishClient(FabricProvider):
     ):
         self.endpoint = endpoint or os.environ.get("SUNFISH_ENDPOINT", DEFAULT_ENDPOINT)
         self.transport = transport
+        self._http = httpx.Client(transport=transport, timeout=30.0)
 
     def _patch(self, body: dict) -> None:
         url = f"http://{self.endpoint}/redfish/v1/Systems/System"
-        with httpx.Client(transport=self.transport, timeout=30.0) as client:
-            resp = client.patch(url, json=body)
+        resp = self._http.patch(url, json=body)
         if resp.status_code not in (200, 204):
             raise FabricError(f"http returned code {resp.status_code}")
 
