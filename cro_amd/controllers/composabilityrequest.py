@@ -162,7 +162,9 @@ class ComposabilityRequestReconciler(Reconciler):
         managed = self.client.list(ComposableResource, {MANAGED_BY_LABEL: request.metadata.name})
         # exclude children already on their way out (:228-235)
         managed = [r for r in managed if r.status.state not in ("Detaching", "Deleting")]
-        all_requests = self.client.list(ComposabilityRequest)
+        # read-only snapshots (never mutated below) — the deep-copy-free
+        # form keeps NodeAllocating O(children) instead of O(fleet)
+        all_requests = self.client.list(ComposabilityRequest, copy=False)
         nodes = get_all_nodes(self.client)
 
         spec = request.spec.resource

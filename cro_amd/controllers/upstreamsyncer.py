@@ -50,7 +50,7 @@ class UpstreamSyncer:
 
         existing_ids = {
             r.status.device_id
-            for r in self.client.list(ComposableResource)
+            for r in self.client.list(ComposableResource, copy=False)  # read-only
             if r.status.device_id
         }
 

@@ -36,7 +36,8 @@ def node_exists(client: Client, name: str) -> bool:
 
 
 def get_all_nodes(client: Client) -> List[Node]:
-    return client.list(Node)
+    # read-only snapshot: allocation only inspects names/capacity
+    return client.list(Node, copy=False)
 
 
 def check_node_capacity_sufficient(

@@ -36,8 +36,16 @@ class Client:
         except NotFoundError:
             return None
 
-    def list(self, cls: Type[T], labels: Optional[Dict[str, str]] = None) -> List[T]:
-        return self.store.list(cls.KIND, labels)
+    def list(
+        self,
+        cls: Type[T],
+        labels: Optional[Dict[str, str]] = None,
+        copy: bool = True,
+    ) -> List[T]:
+        """``copy=False`` = read-only snapshot (no per-object deep copy);
+        see InMemoryStore.list. Remote clients ignore the flag (wire
+        transfer already isolates)."""
+        return self.store.list(cls.KIND, labels, copy=copy)
 
     def update(self, obj: T) -> T:
         return self.store.update(obj)

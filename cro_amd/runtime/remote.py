@@ -132,7 +132,7 @@ class RemoteClient:
         except NotFoundError:
             return None
 
-    def list(self, cls_or_kind, labels: Optional[Dict[str, str]] = None):
+    def list(self, cls_or_kind, labels: Optional[Dict[str, str]] = None, copy: bool = True):
         cls, plural = self._resolve(cls_or_kind)
         params = {}
         if labels:

@@ -190,7 +190,16 @@ class InMemoryStore:
             except KeyError:
                 raise NotFoundError(f"{kind}/{name} not found") from None
 
-    def list(self, kind: str, label_selector: Optional[Dict[str, str]] = None) -> List[K8sObject]:
+    def list(
+        self,
+        kind: str,
+        label_selector: Optional[Dict[str, str]] = None,
+        copy: bool = True,
+    ) -> List[K8sObject]:
+        """``copy=False`` returns the stored objects themselves — READ-ONLY
+        snapshots for hot validation/allocation paths (admission lists every
+        request per CREATE; deep-copying a 400-object fleet per call is the
+        O(n²) term). Callers must not mutate them."""
         with self._lock:
             items = list(self._objects.get(kind, {}).values())
             if label_selector:
@@ -199,6 +208,8 @@ class InMemoryStore:
                     for o in items
                     if all(o.metadata.labels.get(k) == v for k, v in label_selector.items())
                 ]
+            if not copy:
+                return items
             return [o.model_copy(deep=True) for o in items]
 
     def update(self, obj: K8sObject) -> K8sObject:

@@ -82,7 +82,9 @@ def admission_validator(client):
     def _validate(op: str, old, new) -> None:
         if new.spec is None:
             return
-        existing = client.list(ComposabilityRequest)
+        # read-only snapshot: admission runs on every CREATE/UPDATE and a
+        # deep copy of the whole fleet per write is the O(n^2) fleet term
+        existing = client.list(ComposabilityRequest, copy=False)
         msg = validate_composability_request(new, existing)
         if msg:
             raise AdmissionDenied(msg)
