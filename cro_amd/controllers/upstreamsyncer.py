@@ -48,11 +48,18 @@ class UpstreamSyncer:
     def sync(self) -> None:
         device_infos = self.adapter.provider.get_resources()
 
-        existing_ids = {
-            r.status.device_id
-            for r in self.client.list(ComposableResource, copy=False)  # read-only
-            if r.status.device_id
-        }
+        existing_ids = set()
+        for r in self.client.list(ComposableResource, copy=False):  # read-only
+            if r.status.device_id:
+                existing_ids.add(r.status.device_id)
+            # a freshly created detach-CR carries the identity only in its
+            # label until its first reconcile copies it into status — count
+            # it or short sync periods create duplicate detach CRs for the
+            # same orphan (the reference's 1-min period merely hides this
+            # race, upstreamsyncer_controller.go:97-135)
+            label_id = r.metadata.labels.get(READY_TO_DETACH_LABEL, "")
+            if label_id:
+                existing_ids.add(label_id)
 
         for info in device_infos:
             did = info.device_id

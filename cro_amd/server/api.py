@@ -140,6 +140,16 @@ def build_app(client: Client) -> FastAPI:
         from fastapi.responses import StreamingResponse
 
         async def gen():
+            import os
+            import time as _time
+
+            # k8s watch-timeout contract: every stream is closed server-side
+            # after a bounded lifetime and the client reconnects (cheap with
+            # rv resume tokens). Bounds the damage of half-open connections
+            # — a client reading keepalives from a stale server otherwise
+            # never notices it should reconnect.
+            lifetime = float(os.environ.get("CRO_WATCH_TIMEOUT", "300"))
+            stream_deadline = _time.monotonic() + lifetime
             events = client.watch([cls.KIND])
             store = getattr(client, "store", None)
             last_seq = 0
@@ -163,6 +173,8 @@ def build_app(client: Client) -> FastAPI:
                         ) + "\n"
                 loop = asyncio.get_running_loop()
                 while True:
+                    if _time.monotonic() >= stream_deadline:
+                        return  # watch timeout; client resumes by rv token
                     try:
                         ev = await loop.run_in_executor(None, events.get, True, 1.0)
                     except _queue.Empty:

@@ -355,3 +355,98 @@ def test_events_since_foreign_token_forces_relist():
     Client(store).create(make_request("x", target_node="n"))
     assert store.events_since(999) is None
     assert store.events_since(store.current_seq()) == []
+
+
+def test_api_server_state_loss_syncer_repairs(monkeypatch):
+    """Catastrophic apiserver restart with a FRESH store (etcd loss): the
+    operator's watch streams carry now-foreign resume tokens — the server
+    answers Expired + empty re-list — and the fabric still holds the
+    composed device with no CR anywhere. The operator's upstream syncer
+    must repair the orphan through the new server (detach CR → physical
+    detach).
+
+    Short CRO_WATCH_TIMEOUT: in-process uvicorn shutdown can leave the old
+    streams half-open (reading keepalives from the dead store); the k8s
+    watch-timeout forces the reconnect that discovers the new server."""
+    import uvicorn
+
+    monkeypatch.setenv("CRO_WATCH_TIMEOUT", "1.5")
+    port = free_port()
+
+    def start_server(app):
+        server = uvicorn.Server(
+            uvicorn.Config(app, host="127.0.0.1", port=port, log_level="error")
+        )
+        thread = threading.Thread(target=server.run, daemon=True)
+        thread.start()
+        import httpx
+
+        deadline = time.monotonic() + 15
+        while time.monotonic() < deadline:
+            try:
+                if httpx.get(f"http://127.0.0.1:{port}/healthz", timeout=1).status_code == 200:
+                    return server, thread
+            except Exception:
+                time.sleep(0.05)
+        raise AssertionError("server did not come up")
+
+    server_mgr = build_manager(Adapter("DRA", MockFabric()), None)
+    server, thread = start_server(build_app(server_mgr.client))
+
+    remote = RemoteClient(f"http://127.0.0.1:{port}")
+    fabric = MockFabric(models={"mi355x": 8})
+    mgr = build_manager(
+        Adapter("DRA", fabric), None, client=remote, enable_webhook=False,
+        syncer_period=0.3, syncer_grace=0.5,
+    )
+    ops = MockNodeOps(client=mgr.client)
+    mgr.resource_reconciler.node_ops = ops
+    mgr.syncer.node_ops = ops
+
+    orig_add = fabric.add_resource
+
+    def add_resource(resource):
+        did, cdi = orig_add(resource)
+        ops.fabric_composed(resource.spec.target_node, did)
+        return did, cdi
+
+    fabric.add_resource = add_resource
+
+    node = Node()
+    node.metadata.name = "node0"
+    remote.create(node)
+    ops.set_driver("node0", True)
+    mgr.start()
+    try:
+        remote.create(make_request("r1", size=1, target_node="node0"))
+        assert wait_for(
+            lambda: (r := remote.try_get(ComposabilityRequest, "r1")) is not None
+            and r.status.state == "Running",
+            timeout=20,
+        )
+        assert len(fabric.attached_to("node0")) == 1
+
+        # replace the server with a FRESH empty store on the same port
+        server.should_exit = True
+        thread.join(timeout=10)
+        fresh_mgr = build_manager(Adapter("DRA", MockFabric()), None)
+        server, thread = start_server(build_app(fresh_mgr.client))
+        node2 = Node()
+        node2.metadata.name = "node0"
+        remote.create(node2)  # re-register (entrypoint would do this)
+
+        # no CR exists anywhere, yet the fabric holds the device — the
+        # syncer must detect the orphan and walk a full detach through
+        # the new server
+        assert wait_for(lambda: fabric.attached_to("node0") == [], timeout=30), (
+            fabric.attached_to("node0"),
+            [r.metadata.name for r in fresh_mgr.client.list(ComposableResource)],
+        )
+        assert wait_for(
+            lambda: fresh_mgr.client.list(ComposableResource) == [], timeout=20
+        )
+    finally:
+        mgr.stop()
+        remote.close()
+        server.should_exit = True
+        thread.join(timeout=10)
