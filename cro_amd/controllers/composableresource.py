@@ -281,6 +281,18 @@ class ComposableResourceReconciler(Reconciler):
                 self.metrics.fabric_request_seconds.labels(
                     self.adapter.provider.name, "add"
                 ).observe(time.monotonic() - t0)
+            # Fabric sanity: a buggy/poisoned fabric manager handing out a
+            # device some other CR already claims would silently dual-map
+            # one GPU into two workloads (the reference has no such guard)
+            for other in self.client.list(ComposableResource, copy=False):
+                if (
+                    other.metadata.name != resource.metadata.name
+                    and other.status.device_id == device_id
+                ):
+                    raise FabricError(
+                        f"fabric returned device {device_id} already claimed "
+                        f"by {other.metadata.name}; refusing dual attachment"
+                    )
             # Persist the fabric-assigned identity with conflict-retry: a
             # concurrent deletion bumps the RV, and dropping this write
             # would leak the attached device (the CR would reach Deleting

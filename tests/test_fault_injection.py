@@ -163,3 +163,167 @@ def test_concurrent_requests_stress():
             assert stack.fabric.attached_to(f"node{i}") == []
     finally:
         stack.mgr.stop()
+
+
+def test_fabric_double_allocation_refused(mock_world):
+    """A fabric that hands out a device another CR already claims must be
+    refused (dual-mapping one GPU into two workloads), not silently
+    accepted — a guard the reference lacks."""
+    from tests.conftest import drive, make_node, make_request
+    from cro_amd.api.v1alpha1.types import ComposabilityRequest, ComposableResource
+
+    w = mock_world
+    make_node(w.client, "node0")
+    w.ops.set_driver("node0", True)
+    w.client.create(make_request("r1", size=1, target_node="node0"))
+    drive(w.request_rec, "r1")
+    req = w.client.get(ComposabilityRequest, "r1")
+    child1 = next(iter(req.status.resources))
+    drive(w.resource_rec, child1)
+    first = w.client.get(ComposableResource, child1)
+    assert first.status.state == "Online"
+
+    # poison the fabric: always return the already-claimed device
+    dup_id = first.status.device_id
+
+    def poisoned(resource):
+        return dup_id, f"amd.com/gpu={dup_id}"
+
+    w.fabric.add_resource = poisoned
+    w.client.create(make_request("r2", model="mi300x", target_node="node0"))
+    drive(w.request_rec, "r2")
+    req2 = w.client.get(ComposabilityRequest, "r2")
+    child2 = next(iter(req2.status.resources))
+    for _ in range(3):
+        try:
+            w.resource_rec.reconcile(child2)
+        except Exception:
+            pass
+    second = w.client.get(ComposableResource, child2)
+    assert second.status.state != "Online"
+    assert "already claimed" in second.status.error
+    assert second.status.device_id == ""  # identity never persisted
+
+
+def test_flaky_node_exec_lifecycle_converges():
+    """Intermittent node-agent/exec failures (every 3rd call raises) must
+    only slow the lifecycle down, never wedge or corrupt it — the
+    ExecError → status.error → backoff-requeue path, end to end."""
+    from cro_amd.api.v1alpha1.types import ComposabilityRequest, ComposableResource
+    from cro_amd.controllers.composabilityrequest import ComposabilityRequestReconciler
+    from cro_amd.controllers.composableresource import (
+        ComposableResourceReconciler,
+        ReconcileConfig,
+    )
+    from cro_amd.fabric.adapter import Adapter
+    from cro_amd.fabric.mock import MockFabric
+    from cro_amd.nodeops.amdgpu import AmdNodeOps
+    from cro_amd.nodeops.execs import ExecError, MockNodeExec
+    from cro_amd.runtime.client import Client
+    from cro_amd.runtime.store import InMemoryStore
+    from tests.conftest import make_node, make_request
+    from tests.test_nodeops import kfd_fixture
+
+    class FlakyExec:
+        """Seeded 10% failure per call: multi-syscall operations (a KFD
+        enumeration is ~10 reads) still succeed sometimes, so retries can
+        make progress — a deterministic every-Nth injector would starve
+        them forever."""
+
+        def __init__(self, inner, rate=0.10, seed=7):
+            import random
+
+            self.inner = inner
+            self.rng = random.Random(seed)
+            self.rate = rate
+            self.calls = 0
+            self.failures = 0
+
+        def _maybe_fail(self):
+            self.calls += 1
+            if self.rng.random() < self.rate:
+                self.failures += 1
+                raise ExecError("injected agent failure")
+
+        def run(self, *a, **k):
+            self._maybe_fail()
+            return self.inner.run(*a, **k)
+
+        def read_file(self, *a, **k):
+            self._maybe_fail()
+            return self.inner.read_file(*a, **k)
+
+        def write_file(self, *a, **k):
+            self._maybe_fail()
+            return self.inner.write_file(*a, **k)
+
+        def list_dir(self, *a, **k):
+            self._maybe_fail()
+            return self.inner.list_dir(*a, **k)
+
+        def path_exists(self, *a, **k):
+            self._maybe_fail()
+            return self.inner.path_exists(*a, **k)
+
+    client = Client(InMemoryStore())
+    inner = MockNodeExec()
+    ids = kfd_fixture(inner, 1, node="node0")
+    flaky = FlakyExec(inner)
+    ops = AmdNodeOps(
+        flaky, client=client, cdi_dir="/etc/cdi",
+        destructive=False, initially_detached=ids,
+    )
+    ops.enum_cache_ttl = 0.0
+    fabric = MockFabric(bind_inventory=[
+        {"device_id": ids[0], "cdi_device_id": f"amd.com/gpu={ids[0]}", "model": "mi355x"}
+    ])
+    orig_add = fabric.add_resource
+
+    def add(resource):
+        did, cdi = orig_add(resource)
+        ops.simulate_compose(resource.spec.target_node, did)
+        return did, cdi
+
+    fabric.add_resource = add
+    adapter = Adapter("DRA", fabric)
+    res_rec = ComposableResourceReconciler(client, adapter, ops, ReconcileConfig())
+    req_rec = ComposabilityRequestReconciler(client)
+    make_node(client, "node0")
+
+    client.create(make_request("r1", size=1, target_node="node0"))
+
+    def crank(rounds=60):
+        for _ in range(rounds):
+            for req in client.list(ComposabilityRequest):
+                try:
+                    req_rec.reconcile(req.metadata.name)
+                except Exception:
+                    pass
+            for res in client.list(ComposableResource):
+                for rec in (res_rec, req_rec):
+                    try:
+                        rec.reconcile(res.metadata.name)
+                    except Exception:
+                        pass
+            r = client.try_get(ComposabilityRequest, "r1")
+            if r is not None and r.status.state == "Running":
+                return True
+        return False
+
+    assert crank(), client.list(ComposableResource)[0].status
+    assert flaky.failures > 5  # failures actually fired along the way
+
+    client.delete(ComposabilityRequest, "r1")
+    for _ in range(120):
+        for name in [r.metadata.name for r in client.list(ComposabilityRequest)] + [
+            r.metadata.name for r in client.list(ComposableResource)
+        ]:
+            for rec in (req_rec, res_rec):
+                try:
+                    rec.reconcile(name)
+                except Exception:
+                    pass
+        if not client.list(ComposabilityRequest) and not client.list(ComposableResource):
+            break
+    assert client.list(ComposableResource) == []
+    assert fabric.attached_to("node0") == []
