@@ -140,6 +140,7 @@ class EventRecorder:
     # -- internals ---------------------------------------------------------
 
     def _record(self, kind, name, type_: str, reason: str, message: str) -> None:
+        message = message[:2000]  # bound stored size (tracebacks, huge errors)
         digest = hashlib.sha256(
             f"{kind}/{name}/{type_}/{reason}/{message}".encode()
         ).hexdigest()[:12]
