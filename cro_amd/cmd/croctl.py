@@ -81,7 +81,10 @@ def _print_table(plural: str, items: list) -> None:
 
 
 def main(argv=None, client: httpx.Client = None) -> int:
+    from .. import __version__
+
     p = argparse.ArgumentParser(prog="croctl")
+    p.add_argument("--version", action="version", version=f"croctl {__version__}")
     p.add_argument("--server", default=os.environ.get("CRO_SERVER", "http://127.0.0.1:8080"))
     p.add_argument("-o", "--output", choices=["table", "yaml", "json"], default="table")
     p.add_argument("--insecure-skip-tls-verify", action="store_true",

@@ -39,7 +39,10 @@ def parse_port(addr: str, default: int) -> int:
 
 
 def main(argv=None) -> int:
+    from .. import __version__
+
     p = argparse.ArgumentParser(prog="cro-amd-operator")
+    p.add_argument("--version", action="version", version=f"cro-amd-operator {__version__}")
     p.add_argument("--metrics-bind-address", default=":8443")
     p.add_argument("--health-probe-bind-address", default=":8081")
     p.add_argument("--api-bind-address", default=":8080")
