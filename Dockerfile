@@ -6,7 +6,10 @@ FROM rocm/dev-ubuntu-22.04:7.2 AS build
 WORKDIR /src
 COPY cro_amd/ cro_amd/
 RUN /opt/rocm/bin/hipcc --offload-arch=gfx950 -O3 -fPIC -shared \
-      cro_amd/hip/probe.hip -o cro_amd/hip/libcroprobe.so
+      cro_amd/hip/probe.hip -o cro_amd/hip/libcroprobe.so && \
+    /opt/rocm/bin/hipcc --offload-arch=gfx950 -O2 \
+      cro_amd/agent/croagent.cpp -Lcro_amd/hip -lcroprobe \
+      -Wl,-rpath,'$ORIGIN/../hip' -o cro_amd/agent/croagent
 
 FROM rocm/rocm-runtime-ubuntu-22.04:7.2
 RUN useradd -u 65532 -r nonroot
