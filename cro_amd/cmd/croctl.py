@@ -196,7 +196,28 @@ def main(argv=None, client: httpx.Client = None) -> int:
         resp = http.get(f"{BASE}/{args.plural}/{args.name}")
         if resp.status_code != 200:
             return fail(resp)
-        print(yaml.safe_dump(resp.json(), sort_keys=False))
+        obj = resp.json()
+        print(yaml.safe_dump(obj, sort_keys=False))
+        # kubectl-describe parity: the object's recent events
+        ev_resp = http.get(f"{BASE}/events")
+        if ev_resp.status_code == 200:
+            kind = obj.get("kind", "")
+            mine = sorted(
+                (
+                    e for e in ev_resp.json()["items"]
+                    if e.get("involved_kind") == kind
+                    and e.get("involved_name") == args.name
+                ),
+                key=lambda e: e.get("last_seen", ""),
+            )
+            if mine:
+                print("Events:")
+                for e in mine[-15:]:
+                    print(
+                        f"  {(e.get('last_seen') or '')[11:19]}  "
+                        f"{e.get('type', ''):<8} {e.get('reason', ''):<18} "
+                        f"(x{e.get('count', 1)})  {e.get('message', '')[:70]}"
+                    )
         return 0
 
     def update_with_conflict_retry(plural: str, name: str, mutate) -> httpx.Response:

@@ -153,3 +153,17 @@ def test_apply_multidoc_and_directory(cli, tmp_path):
     )
     rc, out, _ = run("apply", "-f", str(d))
     assert rc == 0 and "md3 created" in out
+
+
+def test_describe_includes_events(cli, tmp_path):
+    """kubectl-describe parity: the object's event trail is appended."""
+    run, stack = cli
+    run("apply", "-f", write_request(tmp_path, name="d1"))
+    assert stack.mgr.wait_for(
+        lambda: "Running" in run("get", "composabilityrequests", "d1")[1],
+        timeout=10,
+    )
+    stack.mgr.recorder.flush()
+    rc, out, _ = run("describe", "composabilityrequests", "d1")
+    assert rc == 0
+    assert "Events:" in out and "NodesAllocated" in out and "Running" in out
