@@ -96,6 +96,8 @@ def main(argv=None, client: httpx.Client = None) -> int:
     g = sub.add_parser("get")
     g.add_argument("plural")
     g.add_argument("name", nargs="?")
+    g.add_argument("-l", "--selector", default="",
+                   metavar="K=V[,K2=V2]", help="label selector")
 
     d = sub.add_parser("describe")
     d.add_argument("plural")
@@ -141,7 +143,8 @@ def main(argv=None, client: httpx.Client = None) -> int:
                 return fail(resp)
             items = [resp.json()]
         else:
-            resp = http.get(f"{BASE}/{args.plural}")
+            params = {"labelSelector": args.selector} if args.selector else {}
+            resp = http.get(f"{BASE}/{args.plural}", params=params)
             if resp.status_code != 200:
                 return fail(resp)
             items = resp.json()["items"]

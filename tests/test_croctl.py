@@ -167,3 +167,19 @@ def test_describe_includes_events(cli, tmp_path):
     rc, out, _ = run("describe", "composabilityrequests", "d1")
     assert rc == 0
     assert "Events:" in out and "NodesAllocated" in out and "Running" in out
+
+
+def test_get_with_label_selector(cli, tmp_path):
+    run, stack = cli
+    run("apply", "-f", write_request(tmp_path, name="l1"))
+    assert stack.mgr.wait_for(
+        lambda: "Online" in run("get", "composableresources")[1], timeout=10
+    )
+    rc, out, _ = run(
+        "get", "composableresources", "-l", "app.kubernetes.io/managed-by=l1"
+    )
+    assert rc == 0 and "gpu-" in out
+    rc, out, _ = run(
+        "get", "composableresources", "-l", "app.kubernetes.io/managed-by=ghost"
+    )
+    assert rc == 0 and "gpu-" not in out
