@@ -450,3 +450,51 @@ def test_api_server_state_loss_syncer_repairs(monkeypatch):
         remote.close()
         server.should_exit = True
         thread.join(timeout=10)
+
+
+def test_watch_timeout_recycling_under_churn(api_server, monkeypatch):
+    """With a 1 s watch timeout, streams recycle many times while
+    lifecycles churn; rv-token resumes must make the recycling invisible
+    (no missed events, no stalls)."""
+    monkeypatch.setenv("CRO_WATCH_TIMEOUT", "1")
+    url, server_mgr = api_server
+    remote = RemoteClient(url)
+    fabric = MockFabric(models={"mi355x": 8})
+    mgr = build_manager(Adapter("DRA", fabric), None, client=remote, enable_webhook=False)
+    ops = MockNodeOps(client=mgr.client)
+    mgr.resource_reconciler.node_ops = ops
+
+    orig_add = fabric.add_resource
+
+    def add(resource):
+        did, cdi = orig_add(resource)
+        ops.fabric_composed(resource.spec.target_node, did)
+        return did, cdi
+
+    fabric.add_resource = add
+    node = Node()
+    node.metadata.name = "node0"
+    remote.create(node)
+    ops.set_driver("node0", True)
+    mgr.start()
+    try:
+        t0 = time.monotonic()
+        cycles = 0
+        while time.monotonic() - t0 < 6.0:  # spans ≥5 stream generations
+            name = f"wt-{cycles}"
+            remote.create(make_request(name, size=1, target_node="node0"))
+            assert wait_for(
+                lambda: (r := remote.try_get(ComposabilityRequest, name)) is not None
+                and r.status.state == "Running",
+                timeout=15,
+            ), name
+            remote.delete(ComposabilityRequest, name)
+            assert wait_for(
+                lambda: remote.try_get(ComposabilityRequest, name) is None, timeout=15
+            ), name
+            cycles += 1
+        assert cycles >= 5  # made real progress across recycles
+        assert fabric.attached_to("node0") == []
+    finally:
+        mgr.stop()
+        remote.close()
