@@ -57,6 +57,10 @@ def main(argv=None) -> int:
                    help="serve the API/store without running controllers "
                    "(the apiserver half of a split deployment)")
     p.add_argument("--cdi-dir", default="/etc/cdi")
+    p.add_argument("--data-dir", default="",
+                   help="standalone durability: persist the object store "
+                   "here (atomic snapshots; reload on restart — the etcd "
+                   "analog). Ignored with --api-server.")
     p.add_argument("--destructive", action="store_true",
                    help="perform real PCI remove/rescan and module unload")
     p.add_argument("--max-concurrent-reconciles", type=int, default=8)
@@ -116,9 +120,18 @@ def main(argv=None) -> int:
         from ..runtime.remote import RemoteClient
 
         remote = RemoteClient(args.api_server)
+    store = None
+    if remote is None and args.data_dir:
+        from ..runtime.store import InMemoryStore
+
+        store = InMemoryStore(
+            persist_path=os.path.join(args.data_dir, "state.json")
+        )
+        log.info("durable store at %s", args.data_dir)
     mgr = build_manager(
         adapter,
         None,
+        store=store,
         client=remote,
         max_concurrent_reconciles=args.max_concurrent_reconciles,
         enable_webhook=os.environ.get("ENABLE_WEBHOOKS", "") != "false",
@@ -255,6 +268,8 @@ def main(argv=None) -> int:
 
     server.run()  # serves API + healthz/readyz/metrics until signal
     mgr.stop()
+    if store is not None:
+        store.close()  # final flush of the durable snapshot
     if lock_file is not None:
         lock_file.close()
     return 0

@@ -75,7 +75,18 @@ class _Watcher:
 
 
 class InMemoryStore:
-    def __init__(self) -> None:
+    def __init__(
+        self,
+        persist_path: Optional[str] = None,
+        persist_debounce: float = 0.05,
+    ) -> None:
+        """``persist_path`` enables write-through durability: every
+        mutation schedules a debounced atomic snapshot (tmp + fsync +
+        rename) and a restarted store reloads objects, the RV counter and
+        the watch sequence from it — the standalone analog of etcd (the
+        reference's CRD status IS its checkpoint; without
+        this a standalone restart would forget every request and rely on
+        the syncer to repair orphaned fabric attachments)."""
         self._lock = threading.RLock()
         self._objects: Dict[str, Dict[str, K8sObject]] = {}
         self._rv = itertools.count(1)
@@ -86,6 +97,124 @@ class InMemoryStore:
         # only what it missed instead of a full re-list
         self._event_seq = 0
         self._event_log: "deque[WatchEvent]" = deque(maxlen=4096)
+        self._persist_path = persist_path
+        self._persist_debounce = persist_debounce
+        self._dirty = threading.Event()
+        self._persist_stop = False
+        self._persist_thread: Optional[threading.Thread] = None
+        if persist_path:
+            self._load()
+            self._persist_thread = threading.Thread(
+                target=self._persist_loop, name="store-persist", daemon=True
+            )
+            self._persist_thread.start()
+
+    # -- durability --------------------------------------------------------
+
+    def _load(self) -> None:
+        import json
+        import os
+
+        from ..api.v1alpha1.types import ALL_KINDS
+
+        if not os.path.exists(self._persist_path):
+            return
+        try:
+            with open(self._persist_path) as f:
+                snap = json.load(f)
+        except (OSError, ValueError):
+            import logging
+
+            logging.getLogger(__name__).exception(
+                "store snapshot %s unreadable; starting empty", self._persist_path
+            )
+            return
+        self._rv = itertools.count(int(snap.get("rv", 0)) + 1)
+        # seq continuity keeps pre-restart watch tokens comparable: a
+        # token < seq with an empty log → Expired → clean re-list
+        self._event_seq = int(snap.get("seq", 0))
+        for kind, objs in snap.get("objects", {}).items():
+            cls = ALL_KINDS.get(kind)
+            if cls is None:
+                continue
+            for name, dump in objs.items():
+                try:
+                    self._objects.setdefault(kind, {})[name] = cls.model_validate(dump)
+                except Exception:
+                    import logging
+
+                    logging.getLogger(__name__).exception(
+                        "skipping unreadable %s/%s in snapshot", kind, name
+                    )
+
+    def _snapshot(self) -> dict:
+        with self._lock:
+            return {
+                "rv": self._peek_rv(),
+                "seq": self._event_seq,
+                "objects": {
+                    kind: {
+                        name: obj.model_dump(by_alias=True)
+                        for name, obj in objs.items()
+                    }
+                    for kind, objs in self._objects.items()
+                },
+            }
+
+    def _peek_rv(self) -> int:
+        # itertools.count cannot be peeked; track via a probe without
+        # consuming: replace the counter with one continuing from n+1
+        n = next(self._rv)
+        self._rv = itertools.count(n + 1)
+        return n
+
+    def persist_now(self) -> None:
+        """Synchronous atomic snapshot (graceful shutdown / tests)."""
+        if not self._persist_path:
+            return
+        import json
+        import os
+        import tempfile
+
+        snap = self._snapshot()
+        directory = os.path.dirname(self._persist_path) or "."
+        os.makedirs(directory, exist_ok=True)
+        fd, tmp = tempfile.mkstemp(dir=directory, prefix=".state-")
+        try:
+            with os.fdopen(fd, "w") as f:
+                json.dump(snap, f)
+                f.flush()
+                os.fsync(f.fileno())
+            os.replace(tmp, self._persist_path)
+        except BaseException:
+            try:
+                os.unlink(tmp)
+            except OSError:
+                pass
+            raise
+
+    def _persist_loop(self) -> None:
+        while not self._persist_stop:
+            self._dirty.wait()
+            if self._persist_stop:
+                return
+            time.sleep(self._persist_debounce)  # coalesce write bursts
+            self._dirty.clear()
+            try:
+                self.persist_now()
+            except Exception:  # pragma: no cover - disk trouble
+                import logging
+
+                logging.getLogger(__name__).exception("store snapshot failed")
+
+    def close(self) -> None:
+        """Flush and stop the persistence thread."""
+        if self._persist_path:
+            self._persist_stop = True
+            self._dirty.set()
+            if self._persist_thread is not None:
+                self._persist_thread.join(timeout=2)
+            self.persist_now()
 
     # -- admission ---------------------------------------------------------
 
@@ -147,6 +276,8 @@ class InMemoryStore:
         self._event_log.append(
             WatchEvent(ev.type, ev.object.model_copy(deep=True), None, ev.seq)
         )
+        if self._persist_path:
+            self._dirty.set()
         # deep copies per subscriber so no watcher can mutate shared state
         for w in self._watchers:
             if not w.closed and w.wants(ev.object.kind):
