@@ -182,3 +182,70 @@ def test_tls_flags_must_pair():
 
     with pytest.raises(SystemExit):
         entry(["--tls-cert-file", "/tmp/only-cert.pem"])
+
+
+@pytest.mark.timeout(120)
+def test_data_dir_survives_restart(tmp_path):
+    """--data-dir: SIGTERM the operator, restart it, the fleet state is
+    still there (standalone etcd analog, end to end)."""
+    data = tmp_path / "data"
+    env = dict(os.environ)
+    env.update({"DEVICE_RESOURCE_TYPE": "DRA", "CDI_PROVIDER_TYPE": "MOCK"})
+
+    def start(port):
+        return subprocess.Popen(
+            [
+                sys.executable, "-m", "cro_amd.cmd.main",
+                "--api-bind-address", f":{port}",
+                "--metrics-bind-address", f":{free_port()}",
+                "--serve-only",
+                "--data-dir", str(data),
+            ],
+            cwd=REPO, env=env,
+            stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+        )
+
+    def wait_up(port, proc):
+        deadline = time.monotonic() + 30
+        while time.monotonic() < deadline:
+            try:
+                if httpx.get(f"http://127.0.0.1:{port}/healthz", timeout=1).status_code == 200:
+                    return True
+            except Exception:
+                time.sleep(0.2)
+        raise AssertionError(proc.stdout.read() if proc.poll() is not None else "no healthz")
+
+    port = free_port()
+    proc = start(port)
+    try:
+        wait_up(port, proc)
+        base = f"http://127.0.0.1:{port}/apis/cro.hpsys.ibm.ie.com/v1alpha1"
+        resp = httpx.post(f"{base}/composabilityrequests", json={
+            "apiVersion": "cro.hpsys.ibm.ie.com/v1alpha1",
+            "kind": "ComposabilityRequest",
+            "metadata": {"name": "persist-me"},
+            "spec": {"resource": {"type": "gpu", "model": "mi355x", "size": 2,
+                                  "target_node": "nodeX"}},
+        })
+        assert resp.status_code == 201, resp.text
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+
+    port2 = free_port()
+    proc2 = start(port2)
+    try:
+        wait_up(port2, proc2)
+        base2 = f"http://127.0.0.1:{port2}/apis/cro.hpsys.ibm.ie.com/v1alpha1"
+        resp = httpx.get(f"{base2}/composabilityrequests/persist-me", timeout=5)
+        assert resp.status_code == 200, resp.text
+        assert resp.json()["spec"]["resource"]["size"] == 2
+    finally:
+        proc2.send_signal(signal.SIGTERM)
+        try:
+            proc2.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc2.kill()
