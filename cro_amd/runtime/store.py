@@ -21,9 +21,11 @@ the reconcile semantics depend on:
   source for controllers (watch-driven reconciles are how this build beats
   the reference's 30 s poll quantum).
 
-Unlike etcd-backed apiservers this store is in-memory; persistence across
-operator restarts is delegated to the real cluster in production deployments
-(cro_amd/runtime/client.py keeps the Client surface identical for both).
+In cluster deployments persistence belongs to etcd behind the real
+apiserver (runtime/client.py keeps the Client surface identical).  In
+standalone mode, ``persist_path`` gives this store its own durability:
+debounced atomic snapshots reloaded on restart with RV/watch-sequence
+continuity (the ``--data-dir`` flag).
 """
 
 from __future__ import annotations
