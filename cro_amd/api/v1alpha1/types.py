@@ -242,6 +242,22 @@ class DaemonSet(K8sObject):
     status: DaemonSetStatus = Field(default_factory=DaemonSetStatus)
 
 
+class Machine(K8sObject):
+    """OpenShift Machine (annotation carrier in the FTI node→machine
+    chain). Registered like the reference registers the machine scheme
+    (cmd/main.go:52-59); stored under "namespace/name" keys."""
+
+    KIND: ClassVar[str] = "Machine"
+    apiVersion: str = "machine.openshift.io/v1beta1"
+
+
+class BareMetalHost(K8sObject):
+    """metal3 BareMetalHost (carries cluster-manager.cdi.io/machine)."""
+
+    KIND: ClassVar[str] = "BareMetalHost"
+    apiVersion: str = "metal3.io/v1alpha1"
+
+
 class DeviceConfigDriver(_Model):
     enable: bool = False
     daemonset_name: str = "amd-gpu-driver"
@@ -295,5 +311,7 @@ ALL_KINDS = {
         DaemonSet,
         DeviceConfig,
         Event,
+        Machine,
+        BareMetalHost,
     )
 }

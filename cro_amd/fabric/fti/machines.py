@@ -15,25 +15,13 @@ the cluster-scoped store.
 
 from __future__ import annotations
 
-from typing import ClassVar
-
-from ...api.v1alpha1.types import K8sObject, Node
+from ...api.v1alpha1.types import BareMetalHost, Machine, Node
 from ...runtime.client import Client
 
 MACHINE_ANNOTATION = "machine.openshift.io/machine"
 BMH_ANNOTATION = "metal3.io/BareMetalHost"
 CM_MACHINE_ANNOTATION = "cluster-manager.cdi.io/machine"
 RKE2_PROVIDER_PREFIX = "fsas-cdi://"
-
-
-class Machine(K8sObject):
-    KIND: ClassVar[str] = "Machine"
-    apiVersion: str = "machine.openshift.io/v1beta1"
-
-
-class BareMetalHost(K8sObject):
-    KIND: ClassVar[str] = "BareMetalHost"
-    apiVersion: str = "metal3.io/v1alpha1"
 
 
 class MachineResolutionError(Exception):

@@ -42,6 +42,10 @@ _PLURALS = {cls.KIND: plural for plural, cls in {
     "devicetaintrules": ALL_KINDS["DeviceTaintRule"],
     "nodes": ALL_KINDS["Node"],
     "events": ALL_KINDS["Event"],
+    "machines": ALL_KINDS["Machine"],
+    "baremetalhosts": ALL_KINDS["BareMetalHost"],
+    "deviceconfigs": ALL_KINDS["DeviceConfig"],
+    "daemonsets": ALL_KINDS["DaemonSet"],
 }.items()}
 
 

@@ -21,10 +21,14 @@ from fastapi import FastAPI, HTTPException, Request, Response
 
 from .. import API_VERSION
 from ..api.v1alpha1.types import (
+    BareMetalHost,
     ComposabilityRequest,
     ComposableResource,
+    DaemonSet,
+    DeviceConfig,
     DeviceTaintRule,
     Event,
+    Machine,
     Node,
     ResourceSlice,
 )
@@ -43,6 +47,10 @@ PLURALS = {
     "devicetaintrules": DeviceTaintRule,
     "nodes": Node,
     "events": Event,
+    "machines": Machine,
+    "baremetalhosts": BareMetalHost,
+    "deviceconfigs": DeviceConfig,
+    "daemonsets": DaemonSet,
 }
 
 BASE = "/apis/cro.hpsys.ibm.ie.com/v1alpha1"

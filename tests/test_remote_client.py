@@ -498,3 +498,30 @@ def test_watch_timeout_recycling_under_churn(api_server, monkeypatch):
     finally:
         mgr.stop()
         remote.close()
+
+
+def test_openshift_machine_chain_over_remote(api_server):
+    """FTI's node→Machine→BMH annotation chain resolves through
+    RemoteClient (the cluster shape where the FTI backend runs off-node
+    and reads these objects over the API)."""
+    from cro_amd.api.v1alpha1.types import BareMetalHost, Machine
+    from cro_amd.fabric.fti.machines import resolve_machine_id_openshift
+
+    url, _ = api_server
+    remote = RemoteClient(url)
+    try:
+        node = Node()
+        node.metadata.name = "worker-0"
+        node.metadata.annotations["machine.openshift.io/machine"] = "ns/m-0"
+        remote.create(node)
+        m = Machine()
+        m.metadata.name = "ns/m-0"
+        m.metadata.annotations["metal3.io/BareMetalHost"] = "ns/bmh-0"
+        remote.create(m)
+        bmh = BareMetalHost()
+        bmh.metadata.name = "ns/bmh-0"
+        bmh.metadata.annotations["cluster-manager.cdi.io/machine"] = "uuid-42"
+        remote.create(bmh)
+        assert resolve_machine_id_openshift(remote, "worker-0") == "uuid-42"
+    finally:
+        remote.close()
