@@ -250,7 +250,15 @@ class ComposabilityRequestReconciler(Reconciler):
                 raise ValueError("TargetNode does not meet spec's requirements")
             allocating = [spec.target_node] * to_allocate
         elif spec.allocation_policy == "samenode":
-            if request.status.resources:
+            if samenode_target == "" and request.status.resources:
+                # no surviving child carried the node (children deleted,
+                # stale status entries remain): the implicit target lives
+                # in status (reference resolves it from status.resources)
+                for entry in request.status.resources.values():
+                    if entry.node_name:
+                        samenode_target = entry.node_name
+                        break
+            if request.status.resources and samenode_target:
                 allocating = [samenode_target] * to_allocate
             else:
                 chosen = None

@@ -281,3 +281,38 @@ def test_gc_on_target_node_deleted(mock_world):
     mock_world.request_rec.reconcile("r1")
     got = mock_world.client.get(ComposabilityRequest, "r1")
     assert got.metadata.deletionTimestamp is not None
+
+
+def test_samenode_reallocation_keeps_implicit_node(mock_world):
+    """Re-entering NodeAllocating with stale status entries but no
+    surviving child CRs must keep the implicitly chosen node (resolved
+    from status), never allocate children with an empty target."""
+    from cro_amd.api.v1alpha1.types import ComposabilityRequest, ComposableResource
+    from tests.conftest import drive, make_node, make_request
+
+    w = mock_world
+    make_node(w.client, "node0")
+    make_node(w.client, "node1")
+    w.ops.set_driver("node0", True)
+    w.client.create(make_request("r1", size=1))  # samenode, no target
+    drive(w.request_rec, "r1")
+    req = w.client.get(ComposabilityRequest, "r1")
+    chosen = next(iter(req.status.resources.values())).node_name
+    assert chosen  # a node was picked
+
+    # children vanish out-of-band (GC-like), status entries remain
+    for child in w.client.list(ComposableResource):
+        child.metadata.finalizers = []
+        w.client.update(child)
+        try:
+            w.client.delete(ComposableResource, child.metadata.name)
+        except Exception:
+            pass
+    req = w.client.get(ComposabilityRequest, "r1")
+    req.status.state = "NodeAllocating"
+    w.client.update_status(req)
+    drive(w.request_rec, "r1")
+
+    req = w.client.get(ComposabilityRequest, "r1")
+    targets = {e.node_name for e in req.status.resources.values()}
+    assert targets == {chosen}, targets  # same node, never ""
