@@ -149,3 +149,51 @@ def test_restart_resumes_running_fleet(tmp_path):
     assert mgr2.wait_for(lambda: fabric.attached_to("n0") == [], timeout=15)
     mgr2.stop()
     store2.close()
+
+
+def test_property_machine_reload_matches(tmp_path):
+    """Persistence fuzz: after arbitrary interleavings from the store
+    state machine, close + reload must reproduce the exact object set."""
+    import hypothesis.strategies as st
+    from hypothesis import given, settings
+
+    from tests.test_store_properties import NAMES, make_obj
+
+    @settings(max_examples=25, deadline=None)
+    @given(st.lists(st.tuples(st.sampled_from(["create", "update", "status", "delete"]),
+                              st.sampled_from(NAMES)), max_size=30))
+    def run(ops):
+        path = str(tmp_path / f"fuzz-{hash(tuple(ops)) & 0xffff}.json")
+        if os.path.exists(path):
+            os.unlink(path)
+        store = InMemoryStore(persist_path=path)
+        for op, name in ops:
+            try:
+                if op == "create":
+                    store.create(make_obj(name))
+                elif op == "update":
+                    cur = store.get("ComposableResource", name)
+                    cur.spec.model = "m1"
+                    store.update(cur)
+                elif op == "status":
+                    cur = store.get("ComposableResource", name)
+                    cur.status.state = "Online"
+                    store.update_status(cur)
+                else:
+                    store.delete("ComposableResource", name)
+            except Exception:
+                pass
+        expected = {
+            o.metadata.name: (o.metadata.resourceVersion, o.status.state)
+            for o in store.list("ComposableResource")
+        }
+        store.close()
+        reloaded = InMemoryStore(persist_path=path)
+        got = {
+            o.metadata.name: (o.metadata.resourceVersion, o.status.state)
+            for o in reloaded.list("ComposableResource")
+        }
+        reloaded.close()
+        assert got == expected, (got, expected)
+
+    run()
