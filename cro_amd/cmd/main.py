@@ -130,7 +130,7 @@ def main(argv=None) -> int:
         log.info("durable store at %s", args.data_dir)
     mgr = build_manager(
         adapter,
-        None,
+        None,  # node_ops installed below
         store=store,
         client=remote,
         max_concurrent_reconciles=args.max_concurrent_reconciles,
@@ -138,6 +138,11 @@ def main(argv=None) -> int:
         syncer_period=args.syncer_period,
         metrics_port=parse_port(args.metrics_bind_address, 8443),
     )
+    # FTI/NEC providers resolve Node→machine through the cluster client,
+    # which does not exist until the manager is built — wire it now (a
+    # None client here would crash the first fabric call in production)
+    if getattr(adapter.provider, "client", "n/a") is None:
+        adapter.provider.client = mgr.client
     probe_fn = None
     try:
         from ..nodeops.probe import load_library, probe_fn_for_nodeops

@@ -249,3 +249,48 @@ def test_data_dir_survives_restart(tmp_path):
             proc2.wait(timeout=15)
         except subprocess.TimeoutExpired:
             proc2.kill()
+
+
+@pytest.mark.timeout(120)
+def test_boot_with_fti_fm_backend(tmp_path):
+    """The operator boots with a real fabric backend selected (FTI FM):
+    provider construction + client wiring must not require a live fabric
+    (contacted lazily), and the process serves and shuts down cleanly."""
+    port = free_port()
+    env = dict(os.environ)
+    env.update({
+        "DEVICE_RESOURCE_TYPE": "DRA",
+        "CDI_PROVIDER_TYPE": "FTI_CDI",
+        "FTI_CDI_API_TYPE": "FM",
+        "FTI_CDI_ENDPOINT": "fabric.invalid",
+        "FTI_CDI_TENANT_ID": "t-1",
+    })
+    proc = subprocess.Popen(
+        [
+            sys.executable, "-m", "cro_amd.cmd.main",
+            "--api-bind-address", f":{port}",
+            "--metrics-bind-address", f":{free_port()}",
+            "--syncer-period", "3600",
+            "--cdi-dir", str(tmp_path / "cdi"),
+        ],
+        cwd=REPO, env=env,
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+    )
+    try:
+        deadline = time.monotonic() + 30
+        up = False
+        while time.monotonic() < deadline:
+            try:
+                if httpx.get(f"http://127.0.0.1:{port}/healthz", timeout=1).status_code == 200:
+                    up = True
+                    break
+            except Exception:
+                time.sleep(0.2)
+        assert up, proc.stdout.read() if proc.poll() is not None else "no healthz"
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+    assert proc.returncode == 0
