@@ -129,9 +129,12 @@ class MockNodeExec(NodeExec):
     """
 
     def __init__(self):
+        from collections import deque
+
         self.files: Dict[Tuple[str, str], str] = {}
         self.commands: Dict[tuple, Tuple[int, str, str]] = {}
-        self.calls: List[tuple] = []
+        # bounded: long soaks/benches must not grow the call log unboundedly
+        self.calls: "deque[tuple]" = deque(maxlen=10000)
         self._lock = threading.Lock()
 
     def set_file(self, node: str, path: str, data: str) -> None:
