@@ -300,6 +300,28 @@ class Event(K8sObject):
     source: str = "cro-amd"
 
 
+class LeaseSpec(_Model):
+    """coordination.k8s.io/v1 LeaseSpec field-for-field."""
+
+    holderIdentity: str = ""
+    leaseDurationSeconds: int = 15
+    acquireTime: Optional[str] = None  # RFC3339 micro time
+    renewTime: Optional[str] = None
+    leaseTransitions: int = 0
+
+
+class Lease(K8sObject):
+    """coordination.k8s.io/v1 Lease — leader election (parity with the
+    reference manager's lease-based election, cmd/main.go:137-155:
+    LeaderElectionID c5744f42.hpsys.ibm.ie.com). Held/renewed by
+    runtime.lease.LeaderElector through the normal client surface, so the
+    same election works over the embedded store or the remote API."""
+
+    KIND: ClassVar[str] = "Lease"
+    apiVersion: str = "coordination.k8s.io/v1"
+    spec: LeaseSpec = Field(default_factory=LeaseSpec)
+
+
 ALL_KINDS = {
     cls.KIND: cls
     for cls in (
@@ -311,6 +333,7 @@ ALL_KINDS = {
         DaemonSet,
         DeviceConfig,
         Event,
+        Lease,
         Machine,
         BareMetalHost,
     )

@@ -91,6 +91,8 @@ def main(argv=None, client: httpx.Client = None) -> int:
                    help="accept any server certificate (self-signed demos)")
     p.add_argument("--certificate-authority", default="",
                    metavar="CA_PEM", help="CA bundle for the server certificate")
+    p.add_argument("--token", default=os.environ.get("CRO_API_TOKEN", ""),
+                   help="bearer token for the API (default: $CRO_API_TOKEN)")
     sub = p.add_subparsers(dest="command", required=True)
 
     g = sub.add_parser("get")
@@ -130,7 +132,10 @@ def main(argv=None, client: httpx.Client = None) -> int:
         verify = False
     elif args.certificate_authority:
         verify = args.certificate_authority
-    http = client or httpx.Client(base_url=args.server, timeout=30, verify=verify)
+    headers = {"Authorization": f"Bearer {args.token}"} if args.token else None
+    http = client or httpx.Client(
+        base_url=args.server, timeout=30, verify=verify, headers=headers
+    )
 
     def fail(resp) -> int:
         print(f"error: {resp.status_code}: {resp.text}", file=sys.stderr)

@@ -15,16 +15,24 @@ Env config (composableresource_adapter.go:40-76 and backends):
   NEC_PROVISIONAL_GPU_UUID, SUNFISH_ENDPOINT, ENABLE_WEBHOOKS,
   CRO_FTI_USERNAME/PASSWORD/CLIENT_ID/CLIENT_SECRET/REALM.
 
-Leader election uses an exclusive flock on a lock file — the single-node
-analog of the reference's lease-based election (cmd/main.go:137-155); in a
-multi-replica cluster deployment the lease path belongs to the kube client
-integration.
+Leader election (--leader-elect) uses a coordination.k8s.io Lease held
+through the client surface (runtime/lease.py) — the same protocol as the
+reference's lease-based election (cmd/main.go:137-155, LeaderElectionID
+c5744f42.hpsys.ibm.ie.com). Against a remote API server this excludes
+replicas on different nodes; against the embedded store it is a no-op
+guard (the store is process-local). Lost leadership is fatal: the process
+exits so the standby's controllers never overlap with ours.
+
+Security posture (fail closed): the API (/apis) and node-agent (/agent)
+surfaces require bearer tokens. When CRO_API_TOKEN / CRO_AGENT_TOKEN are
+unset, tokens are auto-generated at startup and written 0600 under
+--data-dir (or /var/run/cro-amd) — a default deployment never exposes an
+unauthenticated write surface.
 """
 
 from __future__ import annotations
 
 import argparse
-import fcntl
 import logging
 import os
 import signal
@@ -46,8 +54,24 @@ def main(argv=None) -> int:
     p.add_argument("--metrics-bind-address", default=":8443")
     p.add_argument("--health-probe-bind-address", default=":8081")
     p.add_argument("--api-bind-address", default=":8080")
-    p.add_argument("--leader-elect", action="store_true")
-    p.add_argument("--leader-elect-lock", default="/var/run/cro-amd/leader.lock")
+    p.add_argument("--leader-elect", action="store_true",
+                   help="gate controllers behind a coordination Lease "
+                   "(holder renews every --leader-retry-period; a standby "
+                   "takes over when renewal stops)")
+    p.add_argument("--leader-lease-duration", type=float, default=15.0)
+    p.add_argument("--leader-renew-deadline", type=float, default=10.0)
+    p.add_argument("--leader-retry-period", type=float, default=2.0)
+    p.add_argument("--webhook-bind-address", default=":9443",
+                   help="AdmissionReview endpoint (ValidatingWebhook"
+                   "Configuration target); serves when ENABLE_WEBHOOKS "
+                   "is not 'false' and TLS material exists")
+    p.add_argument("--webhook-cert-dir",
+                   default="/tmp/k8s-webhook-server/serving-certs",
+                   help="directory with tls.crt/tls.key (cert-manager "
+                   "mount; kubebuilder default path)")
+    p.add_argument("--webhook-insecure", action="store_true",
+                   help="serve the webhook over plain HTTP (tests only; "
+                   "a real apiserver requires TLS)")
     p.add_argument("--node", default=os.environ.get("NODE_NAME", ""),
                    help="local node name this operator instance manages")
     p.add_argument("--api-server", default="",
@@ -100,13 +124,40 @@ def main(argv=None) -> int:
         )
     log = logging.getLogger("cro_amd.main")
 
-    lock_file = None
-    if args.leader_elect:
-        os.makedirs(os.path.dirname(args.leader_elect_lock), exist_ok=True)
-        lock_file = open(args.leader_elect_lock, "w")
-        log.info("waiting for leader lock %s", args.leader_elect_lock)
-        fcntl.flock(lock_file, fcntl.LOCK_EX)
-        log.info("acquired leadership")
+    # -- fail-closed bearer tokens ------------------------------------------
+    # Deployments set these from the cro-amd-tokens Secret; standalone runs
+    # get fresh random tokens written 0600 so nothing serves unauthenticated.
+    token_dir = args.data_dir or "/var/run/cro-amd"
+    api_token = os.environ.get("CRO_API_TOKEN", "")
+    agent_token = os.environ.get("CRO_AGENT_TOKEN", "")
+    generated = {}
+    if not api_token:
+        import secrets
+
+        api_token = secrets.token_hex(24)
+        generated["api.token"] = api_token
+    if not agent_token and args.node:
+        import secrets
+
+        agent_token = secrets.token_hex(24)
+        generated["agent.token"] = agent_token
+    if generated:
+        try:
+            os.makedirs(token_dir, exist_ok=True)
+            for fname, value in generated.items():
+                path = os.path.join(token_dir, fname)
+                fd = os.open(path, os.O_WRONLY | os.O_CREAT | os.O_TRUNC, 0o600)
+                with os.fdopen(fd, "w") as f:
+                    f.write(value)
+                log.warning(
+                    "no %s configured; generated one at %s (set the env var "
+                    "from a Secret for multi-process deployments)",
+                    "CRO_API_TOKEN" if fname == "api.token" else "CRO_AGENT_TOKEN",
+                    path,
+                )
+        except OSError as exc:
+            log.warning("could not persist generated tokens (%s); clients "
+                        "of this process must read them from the log", exc)
 
     from ..controllers import build_manager
     from ..fabric.adapter import new_adapter
@@ -235,21 +286,16 @@ def main(argv=None) -> int:
             adapter.provider = fabric
             log.info("MOCK fabric bound to %d local device(s)", len(gpus))
 
-    if not args.serve_only:
-        mgr.start()
-        log.info("manager started (%d reconcile workers per controller)",
-                 args.max_concurrent_reconciles)
-    else:
-        log.info("serve-only mode: API/store up, controllers disabled")
-
     from ..server.api import build_app
     from ..server.agent_api import build_agent_app
 
     import uvicorn
 
-    app = build_app(mgr.client)
+    app = build_app(mgr.client, token=api_token)
     if args.node:  # node-agent surface for off-node controllers
-        build_agent_app(node_ops.execer, node_name=args.node, app=app)
+        build_agent_app(
+            node_ops.execer, node_name=args.node, app=app, token=agent_token
+        )
     server = uvicorn.Server(
         uvicorn.Config(
             app,
@@ -271,12 +317,98 @@ def main(argv=None) -> int:
     signal.signal(signal.SIGTERM, handle_signal)
     signal.signal(signal.SIGINT, handle_signal)
 
+    # -- admission webhook listener (:9443) ---------------------------------
+    # The reference serves its ValidatingWebhook from the manager process
+    # (cmd/main.go:196-201); config/webhook/manifests.yaml registers this
+    # endpoint with failurePolicy=Fail, so the listener MUST exist whenever
+    # the manifests are applied.
+    webhook_server = None
+    webhook_thread = None
+    enable_webhooks = os.environ.get("ENABLE_WEBHOOKS", "") != "false"
+    if enable_webhooks:
+        cert = os.path.join(args.webhook_cert_dir, "tls.crt")
+        key = os.path.join(args.webhook_cert_dir, "tls.key")
+        have_tls = os.path.exists(cert) and os.path.exists(key)
+        if have_tls or args.webhook_insecure:
+            from ..api.v1alpha1.types import ComposabilityRequest
+            from ..webhook.server import build_app as build_webhook_app
+
+            webhook_app = build_webhook_app(
+                lambda: mgr.client.list(ComposabilityRequest)
+            )
+            webhook_server = uvicorn.Server(
+                uvicorn.Config(
+                    webhook_app,
+                    host="0.0.0.0",
+                    port=parse_port(args.webhook_bind_address, 9443),
+                    log_level="warning",
+                    ssl_certfile=cert if have_tls else None,
+                    ssl_keyfile=key if have_tls else None,
+                )
+            )
+            webhook_thread = threading.Thread(
+                target=webhook_server.run, name="webhook-server", daemon=True
+            )
+            webhook_thread.start()
+            log.info(
+                "admission webhook serving on %s (%s)",
+                args.webhook_bind_address,
+                "TLS" if have_tls else "INSECURE plain HTTP",
+            )
+        else:
+            log.warning(
+                "ENABLE_WEBHOOKS is on but %s has no tls.crt/tls.key; "
+                "NOT serving the AdmissionReview endpoint — do not apply "
+                "config/webhook/manifests.yaml without certificates",
+                args.webhook_cert_dir,
+            )
+
+    # -- leader election + controller start ---------------------------------
+    elector = None
+    if not args.serve_only:
+        def start_controllers():
+            mgr.start()
+            log.info("manager started (%d reconcile workers per controller)",
+                     args.max_concurrent_reconciles)
+
+        if args.leader_elect:
+            from ..runtime.lease import LeaderElector
+
+            import socket as _socket
+
+            def lost_leadership():
+                # fatal, as in controller-runtime: never run controllers
+                # concurrently with the new leader
+                log.error("leadership lost; shutting down")
+                server.should_exit = True
+                stop.set()
+
+            elector = LeaderElector(
+                mgr.client,
+                identity=f"{_socket.gethostname()}_{os.getpid()}",
+                lease_duration=args.leader_lease_duration,
+                renew_deadline=args.leader_renew_deadline,
+                retry_period=args.leader_retry_period,
+                on_started_leading=start_controllers,
+                on_stopped_leading=lost_leadership,
+            )
+            elector.start()
+            log.info("waiting for leader lease as %s", elector.identity)
+        else:
+            start_controllers()
+    else:
+        log.info("serve-only mode: API/store up, controllers disabled")
+
     server.run()  # serves API + healthz/readyz/metrics until signal
+    if elector is not None:
+        elector.stop()  # releases the lease for fast standby takeover
+    if webhook_server is not None:
+        webhook_server.should_exit = True
+        if webhook_thread is not None:
+            webhook_thread.join(timeout=5)
     mgr.stop()
     if store is not None:
         store.close()  # final flush of the durable snapshot
-    if lock_file is not None:
-        lock_file.close()
     return 0
 
 

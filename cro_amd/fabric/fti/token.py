@@ -78,19 +78,30 @@ class CachedToken:
         self.verify = verify
         self.leeway = EXPIRY_LEEWAY
         self._lock = threading.Lock()
-        self._token: Optional[str] = None
-        self._expiry: float = 0.0
+        # (token, expiry) kept as ONE tuple so the unlocked fast-path read
+        # is atomic — two separate attribute reads could pair an old token
+        # with a refreshed expiry under a concurrent refresh
+        self._cached: "tuple[Optional[str], float]" = (None, 0.0)
+
+    @property
+    def _token(self) -> Optional[str]:
+        return self._cached[0]
+
+    @property
+    def _expiry(self) -> float:
+        return self._cached[1]
 
     def get_token(self) -> str:
         now = time.time()
-        token, expiry = self._token, self._expiry
+        token, expiry = self._cached
         if token is not None and expiry - self.leeway > now:
             return token
         with self._lock:
-            if self._token is not None and self._expiry - self.leeway > now:
-                return self._token
-            self._token, self._expiry = self._fetch()
-            return self._token
+            token, expiry = self._cached
+            if token is not None and expiry - self.leeway > now:
+                return token
+            self._cached = self._fetch()
+            return self._cached[0]
 
     def _fetch(self) -> "tuple[str, float]":
         creds = self.credentials()

@@ -42,6 +42,7 @@ _PLURALS = {cls.KIND: plural for plural, cls in {
     "devicetaintrules": ALL_KINDS["DeviceTaintRule"],
     "nodes": ALL_KINDS["Node"],
     "events": ALL_KINDS["Event"],
+    "leases": ALL_KINDS["Lease"],
     "machines": ALL_KINDS["Machine"],
     "baremetalhosts": ALL_KINDS["BareMetalHost"],
     "deviceconfigs": ALL_KINDS["DeviceConfig"],

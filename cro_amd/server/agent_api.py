@@ -19,6 +19,13 @@ Authentication: set ``CRO_AGENT_TOKEN`` (or pass ``token=``) and every
 analog of the RBAC that gated the reference's pods/exec path
 (config/rbac/role.yaml pods/exec verbs). RemoteNodeExec sends the token
 from the same env var.
+
+FAIL-CLOSED: building the agent surface with no token configured raises
+``AgentAuthError`` — these routes execute binaries and write files as the
+operator user, so an unauthenticated surface bound to 0.0.0.0 would be
+remote code execution. Tests and loopback-only deployments may opt out
+explicitly with ``allow_insecure=True``; the production entrypoint
+auto-generates a token instead (cmd/main.py).
 """
 
 from __future__ import annotations
@@ -31,14 +38,28 @@ from fastapi import FastAPI, HTTPException, Request, Response
 from ..nodeops.execs import ExecError, NodeExec
 
 
+class AgentAuthError(RuntimeError):
+    """No agent token configured and insecure mode not explicitly allowed."""
+
+
 def build_agent_app(
     execer: NodeExec, node_name: str = "local", app: FastAPI = None,
-    token: str = None,
+    token: str = None, allow_insecure: bool = False,
 ) -> FastAPI:
     """Build the agent app, or graft the /agent routes onto an existing app
-    (the operator entrypoint serves API + agent surface in one process)."""
+    (the operator entrypoint serves API + agent surface in one process).
+
+    Refuses to register the routes when no bearer token is configured
+    (fail closed — see module docstring) unless ``allow_insecure=True``.
+    """
     if token is None:
         token = os.environ.get("CRO_AGENT_TOKEN", "")
+    if not token and not allow_insecure:
+        raise AgentAuthError(
+            "refusing to serve the /agent API without a bearer token: set "
+            "CRO_AGENT_TOKEN (or pass token=), or pass allow_insecure=True "
+            "for loopback-only test deployments"
+        )
 
     def authorize(request: Request) -> None:
         if not token:

@@ -28,6 +28,7 @@ from ..api.v1alpha1.types import (
     DeviceConfig,
     DeviceTaintRule,
     Event,
+    Lease,
     Machine,
     Node,
     ResourceSlice,
@@ -47,6 +48,7 @@ PLURALS = {
     "devicetaintrules": DeviceTaintRule,
     "nodes": Node,
     "events": Event,
+    "leases": Lease,
     "machines": Machine,
     "baremetalhosts": BareMetalHost,
     "deviceconfigs": DeviceConfig,
@@ -84,8 +86,45 @@ def _http_error(exc: Exception) -> HTTPException:
     return HTTPException(500, {"reason": "InternalError", "message": str(exc)})
 
 
-def build_app(client: Client) -> FastAPI:
+def build_app(client: Client, token: str = None) -> FastAPI:
+    """apiserver-shaped app over ``client``.
+
+    ``token`` (default: env ``CRO_API_TOKEN``) gates every ``/apis`` route
+    with ``Authorization: Bearer`` — the standalone analog of the
+    kube-apiserver's authn in front of the CRD API. The production
+    entrypoint auto-generates one when unset (cmd/main.py), so a default
+    split deployment is never an open write surface to PCI remove/rescan
+    (--destructive). Health and readiness stay unauthenticated, matching
+    kubelet probe semantics; /metrics has its own token (below).
+    """
+    import hmac as _hmac
+    import os as _os
+
+    if token is None:
+        token = _os.environ.get("CRO_API_TOKEN", "")
+
     app = FastAPI(title="cro-amd API", version=API_VERSION)
+
+    if token:
+        @app.middleware("http")
+        async def _api_auth(request: Request, call_next):
+            if request.url.path.startswith("/apis"):
+                auth = request.headers.get("authorization", "")
+                if not _hmac.compare_digest(auth, f"Bearer {token}"):
+                    from fastapi.responses import JSONResponse
+
+                    return JSONResponse(
+                        status_code=401,
+                        content={
+                            "kind": "Status",
+                            "apiVersion": "v1",
+                            "status": "Failure",
+                            "reason": "Unauthorized",
+                            "message": "API requires a valid bearer token",
+                            "code": 401,
+                        },
+                    )
+            return await call_next(request)
 
     @app.get("/healthz")
     def healthz():

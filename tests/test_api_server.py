@@ -119,3 +119,35 @@ def test_metrics_token_auth(api_stack, monkeypatch):
     assert http.get("/metrics", headers={"Authorization": "Bearer wrong"}).status_code == 401
     ok = http.get("/metrics", headers={"Authorization": "Bearer sekrit"})
     assert ok.status_code == 200 and b"cro_reconcile_total" in ok.content
+
+
+def test_api_token_enforced_on_apis_routes():
+    """CRO_API_TOKEN (or token=) gates every /apis route with a k8s
+    Status-shaped 401; /healthz+/readyz stay open for kubelet probes."""
+    from fastapi.testclient import TestClient
+
+    from cro_amd.controllers import build_manager
+    from cro_amd.fabric.adapter import Adapter
+    from cro_amd.fabric.mock import MockFabric
+    from cro_amd.server.api import build_app
+
+    mgr = build_manager(Adapter("DRA", MockFabric()), None, enable_webhook=False)
+    app = build_app(mgr.client, token="sekrit")
+    c = TestClient(app)
+
+    r = c.get("/apis/cro.hpsys.ibm.ie.com/v1alpha1/nodes")
+    assert r.status_code == 401
+    assert r.json()["kind"] == "Status"
+    assert r.json()["reason"] == "Unauthorized"
+    r = c.post("/apis/cro.hpsys.ibm.ie.com/v1alpha1/nodes", json={
+        "apiVersion": "v1", "kind": "Node", "metadata": {"name": "n"}})
+    assert r.status_code == 401
+    # health probes stay open
+    assert c.get("/healthz").status_code == 200
+    assert c.get("/readyz").status_code == 200
+    # correct token passes
+    ok = {"Authorization": "Bearer sekrit"}
+    assert c.get("/apis/cro.hpsys.ibm.ie.com/v1alpha1/nodes", headers=ok).status_code == 200
+    # wrong token still refused
+    bad = {"Authorization": "Bearer wrong"}
+    assert c.get("/apis/cro.hpsys.ibm.ie.com/v1alpha1/nodes", headers=bad).status_code == 401

@@ -27,7 +27,10 @@ def free_port() -> int:
 def test_main_serves_and_shuts_down(tmp_path):
     port = free_port()
     env = dict(os.environ)
-    env.update({"DEVICE_RESOURCE_TYPE": "DRA", "CDI_PROVIDER_TYPE": "MOCK"})
+    env.update({
+        "DEVICE_RESOURCE_TYPE": "DRA", "CDI_PROVIDER_TYPE": "MOCK",
+        "CRO_API_TOKEN": "test-api-token", "CRO_AGENT_TOKEN": "test-agent-token",
+    })
     proc = subprocess.Popen(
         [
             sys.executable, "-m", "cro_amd.cmd.main",
@@ -35,7 +38,8 @@ def test_main_serves_and_shuts_down(tmp_path):
             "--metrics-bind-address", f":{free_port()}",
             "--node", "test-node",
             "--cdi-dir", str(tmp_path / "cdi"),
-            "--leader-elect", "--leader-elect-lock", str(tmp_path / "leader.lock"),
+            "--leader-elect",
+            "--data-dir", str(tmp_path / "data"),
         ],
         cwd=REPO,
         env=env,
@@ -56,9 +60,17 @@ def test_main_serves_and_shuts_down(tmp_path):
                 time.sleep(0.2)
         assert up, proc.stdout.read() if proc.poll() is not None else "no healthz"
 
-        # API surface is live
+        # API surface is fail-closed: no token → 401 k8s Status
         resp = httpx.get(
             base + "/apis/cro.hpsys.ibm.ie.com/v1alpha1/composabilityrequests",
+            timeout=5,
+        )
+        assert resp.status_code == 401
+        assert resp.json()["reason"] == "Unauthorized"
+        # ... and live with the bearer token
+        resp = httpx.get(
+            base + "/apis/cro.hpsys.ibm.ie.com/v1alpha1/composabilityrequests",
+            headers={"Authorization": "Bearer test-api-token"},
             timeout=5,
         )
         assert resp.status_code == 200
@@ -106,17 +118,51 @@ def test_operator_restart_resumes_state_machine():
         mgr2.stop()
 
 
-def test_leader_election_flock(tmp_path):
-    import fcntl
-
-    lock_path = tmp_path / "leader.lock"
-    first = open(lock_path, "w")
-    fcntl.flock(first, fcntl.LOCK_EX)
-    second = open(lock_path, "w")
-    with pytest.raises(BlockingIOError):
-        fcntl.flock(second, fcntl.LOCK_EX | fcntl.LOCK_NB)
-    fcntl.flock(first, fcntl.LOCK_UN)
-    fcntl.flock(second, fcntl.LOCK_EX | fcntl.LOCK_NB)  # now acquirable
+def test_token_autogeneration(tmp_path):
+    """No CRO_API_TOKEN in the env → the entrypoint generates one, persists
+    it 0600 under --data-dir, and the API refuses unauthenticated access
+    while accepting the generated token (fail-closed default)."""
+    port = free_port()
+    data = tmp_path / "data"
+    env = {k: v for k, v in os.environ.items()
+           if k not in ("CRO_API_TOKEN", "CRO_AGENT_TOKEN")}
+    env.update({"DEVICE_RESOURCE_TYPE": "DRA", "CDI_PROVIDER_TYPE": "MOCK"})
+    proc = subprocess.Popen(
+        [
+            sys.executable, "-m", "cro_amd.cmd.main",
+            "--api-bind-address", f":{port}",
+            "--metrics-bind-address", f":{free_port()}",
+            "--serve-only",
+            "--data-dir", str(data),
+        ],
+        cwd=REPO, env=env,
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+    )
+    try:
+        base = f"http://127.0.0.1:{port}"
+        deadline = time.monotonic() + 30
+        while time.monotonic() < deadline:
+            try:
+                if httpx.get(base + "/healthz", timeout=1).status_code == 200:
+                    break
+            except Exception:
+                time.sleep(0.2)
+        else:
+            raise AssertionError(proc.stdout.read() if proc.poll() is not None else "no healthz")
+        url = base + "/apis/cro.hpsys.ibm.ie.com/v1alpha1/nodes"
+        assert httpx.get(url, timeout=5).status_code == 401
+        token_path = data / "api.token"
+        assert token_path.exists()
+        assert (token_path.stat().st_mode & 0o777) == 0o600
+        token = token_path.read_text().strip()
+        resp = httpx.get(url, headers={"Authorization": f"Bearer {token}"}, timeout=5)
+        assert resp.status_code == 200
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
 
 
 @pytest.mark.timeout(120)
@@ -135,7 +181,8 @@ def test_main_serves_tls(tmp_path):
     )
     port = free_port()
     env = dict(os.environ)
-    env.update({"DEVICE_RESOURCE_TYPE": "DRA", "CDI_PROVIDER_TYPE": "MOCK"})
+    env.update({"DEVICE_RESOURCE_TYPE": "DRA", "CDI_PROVIDER_TYPE": "MOCK",
+                "CRO_API_TOKEN": "tls-tok"})
     proc = subprocess.Popen(
         [
             sys.executable, "-m", "cro_amd.cmd.main",
@@ -166,9 +213,9 @@ def test_main_serves_tls(tmp_path):
         from cro_amd.cmd.croctl import main as croctl
 
         assert croctl(["--server", base, "--insecure-skip-tls-verify",
-                       "get", "nodes"]) == 0
+                       "--token", "tls-tok", "get", "nodes"]) == 0
         assert croctl(["--server", base, "--certificate-authority", str(cert),
-                       "get", "nodes"]) == 0
+                       "--token", "tls-tok", "get", "nodes"]) == 0
     finally:
         proc.send_signal(signal.SIGTERM)
         try:
@@ -190,7 +237,9 @@ def test_data_dir_survives_restart(tmp_path):
     still there (standalone etcd analog, end to end)."""
     data = tmp_path / "data"
     env = dict(os.environ)
-    env.update({"DEVICE_RESOURCE_TYPE": "DRA", "CDI_PROVIDER_TYPE": "MOCK"})
+    env.update({"DEVICE_RESOURCE_TYPE": "DRA", "CDI_PROVIDER_TYPE": "MOCK",
+                "CRO_API_TOKEN": "persist-tok"})
+    auth = {"Authorization": "Bearer persist-tok"}
 
     def start(port):
         return subprocess.Popen(
@@ -220,7 +269,7 @@ def test_data_dir_survives_restart(tmp_path):
     try:
         wait_up(port, proc)
         base = f"http://127.0.0.1:{port}/apis/cro.hpsys.ibm.ie.com/v1alpha1"
-        resp = httpx.post(f"{base}/composabilityrequests", json={
+        resp = httpx.post(f"{base}/composabilityrequests", headers=auth, json={
             "apiVersion": "cro.hpsys.ibm.ie.com/v1alpha1",
             "kind": "ComposabilityRequest",
             "metadata": {"name": "persist-me"},
@@ -240,7 +289,8 @@ def test_data_dir_survives_restart(tmp_path):
     try:
         wait_up(port2, proc2)
         base2 = f"http://127.0.0.1:{port2}/apis/cro.hpsys.ibm.ie.com/v1alpha1"
-        resp = httpx.get(f"{base2}/composabilityrequests/persist-me", timeout=5)
+        resp = httpx.get(f"{base2}/composabilityrequests/persist-me",
+                         headers=auth, timeout=5)
         assert resp.status_code == 200, resp.text
         assert resp.json()["spec"]["resource"]["size"] == 2
     finally:
@@ -249,6 +299,96 @@ def test_data_dir_survives_restart(tmp_path):
             proc2.wait(timeout=15)
         except subprocess.TimeoutExpired:
             proc2.kill()
+
+
+@pytest.mark.timeout(120)
+def test_webhook_served_from_entrypoint(tmp_path):
+    """The production entrypoint serves the AdmissionReview endpoint on
+    :9443 with the cert-dir TLS material (cmd/main.go:196-201 parity) —
+    a CREATE violating webhook rule 1 is rejected THROUGH the HTTPS
+    endpoint that config/webhook/manifests.yaml registers."""
+    certdir = tmp_path / "serving-certs"
+    certdir.mkdir()
+    subprocess.run(
+        ["openssl", "req", "-x509", "-newkey", "rsa:2048", "-nodes",
+         "-keyout", str(certdir / "tls.key"), "-out", str(certdir / "tls.crt"),
+         "-days", "1", "-subj", "/CN=127.0.0.1",
+         "-addext", "subjectAltName=IP:127.0.0.1"],
+        check=True, capture_output=True,
+    )
+    api_port, wh_port = free_port(), free_port()
+    env = dict(os.environ)
+    env.update({"DEVICE_RESOURCE_TYPE": "DRA", "CDI_PROVIDER_TYPE": "MOCK",
+                "CRO_API_TOKEN": "wh-tok", "ENABLE_WEBHOOKS": "true"})
+    proc = subprocess.Popen(
+        [
+            sys.executable, "-m", "cro_amd.cmd.main",
+            "--api-bind-address", f":{api_port}",
+            "--metrics-bind-address", f":{free_port()}",
+            "--webhook-bind-address", f":{wh_port}",
+            "--webhook-cert-dir", str(certdir),
+            "--serve-only",
+        ],
+        cwd=REPO, env=env,
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+    )
+    try:
+        base = f"https://127.0.0.1:{wh_port}"
+        deadline = time.monotonic() + 30
+        while time.monotonic() < deadline:
+            try:
+                if httpx.get(base + "/healthz", timeout=1,
+                             verify=str(certdir / "tls.crt")).status_code == 200:
+                    break
+            except Exception:
+                time.sleep(0.2)
+        else:
+            raise AssertionError(
+                proc.stdout.read() if proc.poll() is not None else "no webhook healthz")
+
+        review = {
+            "apiVersion": "admission.k8s.io/v1",
+            "kind": "AdmissionReview",
+            "request": {
+                "uid": "test-uid-1",
+                "operation": "CREATE",
+                "object": {
+                    "apiVersion": "cro.hpsys.ibm.ie.com/v1alpha1",
+                    "kind": "ComposabilityRequest",
+                    "metadata": {"name": "bad"},
+                    "spec": {"resource": {
+                        "type": "gpu", "model": "mi355x", "size": 1,
+                        "allocation_policy": "differentnode",
+                        "target_node": "nodeX",  # rule 1 violation
+                    }},
+                },
+            },
+        }
+        resp = httpx.post(
+            base + "/validate-cro-hpsys-ibm-ie-com-v1alpha1-composabilityrequest",
+            json=review, timeout=5, verify=str(certdir / "tls.crt"),
+        )
+        assert resp.status_code == 200
+        body = resp.json()
+        assert body["response"]["uid"] == "test-uid-1"
+        assert body["response"]["allowed"] is False
+        assert "differentnode" in body["response"]["status"]["message"]
+
+        # a valid CREATE passes through the same endpoint
+        review["request"]["uid"] = "test-uid-2"
+        review["request"]["object"]["spec"]["resource"].pop("target_node")
+        resp = httpx.post(
+            base + "/validate-cro-hpsys-ibm-ie-com-v1alpha1-composabilityrequest",
+            json=review, timeout=5, verify=str(certdir / "tls.crt"),
+        )
+        assert resp.json()["response"]["allowed"] is True
+    finally:
+        proc.send_signal(signal.SIGTERM)
+        try:
+            proc.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            proc.kill()
+    assert proc.returncode == 0
 
 
 @pytest.mark.timeout(120)

@@ -23,7 +23,7 @@ def remote_exec():
     """Agent serving a MockNodeExec-backed node, reached over ASGI HTTP."""
     backend = MockNodeExec()
     ids = kfd_fixture(backend, 2, node=NODE)
-    app = build_agent_app(backend, node_name=NODE)
+    app = build_agent_app(backend, node_name=NODE, allow_insecure=True)
     http = TestClient(app)
 
     class TestClientTransport:
@@ -90,7 +90,7 @@ def test_resolver_miss():
 def test_agent_api_against_real_fs(tmp_path):
     """LocalNodeExec with a sysroot behind the agent API: file IO hits disk."""
     os.makedirs(tmp_path / "sys" / "module" / "amdgpu")
-    app = build_agent_app(LocalNodeExec(sysroot=str(tmp_path)), node_name="n")
+    app = build_agent_app(LocalNodeExec(sysroot=str(tmp_path)), node_name="n", allow_insecure=True)
     http = TestClient(app)
     remote = RemoteNodeExec({"n": "http://agent"})
     remote._http = http
@@ -155,3 +155,14 @@ def test_agent_token_auth():
     wrong._http = TestClient(app, headers={"Authorization": "Bearer nope"})
     with pytest.raises(Exception):
         wrong.read_file(NODE, "/sys/class/kfd/kfd/topology/nodes/1/properties")
+
+
+def test_agent_app_fails_closed_without_token(monkeypatch):
+    """No token and no explicit insecure opt-in → the /agent surface must
+    refuse to build (it executes binaries and writes files)."""
+    from cro_amd.server.agent_api import AgentAuthError
+
+    monkeypatch.delenv("CRO_AGENT_TOKEN", raising=False)
+    backend = MockNodeExec()
+    with pytest.raises(AgentAuthError):
+        build_agent_app(backend, node_name=NODE)
