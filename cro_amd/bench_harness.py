@@ -57,6 +57,8 @@ def build_local_stack(
     mode: str = "DRA",
     use_gpu: Optional[bool] = None,
     gpu_index: Optional[int] = None,
+    gpu_count: Optional[int] = None,
+    mock_devices: int = 8,
     fabric_config: Optional[MockFabricConfig] = None,
     enable_probe: bool = True,
     max_concurrent_reconciles: int = 8,
@@ -78,6 +80,8 @@ def build_local_stack(
             raise RuntimeError("no GPUs in KFD topology")
         if gpu_index is not None:
             gpus = [gpus[gpu_index % len(gpus)]]
+        elif gpu_count is not None:
+            gpus = gpus[:gpu_count]
         device_ids = [g.device_id for g in gpus]
         probe_fn = None
         if enable_probe:
@@ -126,7 +130,7 @@ def build_local_stack(
 
         fabric.add_resource = add_resource
     else:
-        fabric = MockFabric(config=fabric_config, models={"mi355x": 8})
+        fabric = MockFabric(config=fabric_config, models={"mi355x": mock_devices})
         adapter = Adapter(mode, fabric)
         mgr = build_manager(
             adapter,
@@ -171,11 +175,20 @@ def build_local_stack(
     return stack
 
 
-def attach_detach_cycle(stack: LocalStack, name: str, size: int = 1, timeout: float = 60.0, force_detach: bool = False) -> dict:
+def attach_detach_cycle(
+    stack: LocalStack, name: str, size: int = 1, timeout: float = 60.0,
+    force_detach: bool = False, events=None,
+) -> dict:
     """One full ComposabilityRequest lifecycle; returns timing samples.
 
     attach_ms = create → Running (every device Online with CDI written);
     detach_ms = delete → object gone (device drained + fabric detach done).
+
+    ``events``: a dedicated ComposabilityRequest watch queue for this
+    caller.  Concurrent driver threads MUST each pass their own
+    (``stack.mgr.store.watch([...])``) — the shared ``stack.request_events``
+    fallback is single-consumer and threads would steal each other's
+    events from it.
     """
     import queue as _queue
 
@@ -190,7 +203,8 @@ def attach_detach_cycle(stack: LocalStack, name: str, size: int = 1, timeout: fl
     )
     req.metadata.name = name
 
-    events = getattr(stack, "request_events", None)
+    if events is None:
+        events = getattr(stack, "request_events", None)
 
     def wait_event(pred) -> bool:
         """Event-driven wait on the stack's request watch (no poll
@@ -253,6 +267,83 @@ def attach_detach_cycle(stack: LocalStack, name: str, size: int = 1, timeout: fl
             f"(state={cur.status.state if cur else 'gone'}, children={children})"
         )
     return {"attach_ms": (t1 - t0) * 1e3, "detach_ms": (t2 - t1) * 1e3}
+
+
+def churn_cycle(
+    stack: LocalStack, name: str, sizes=(1, 4, 8, 0), timeout: float = 120.0,
+    force_detach: bool = False,
+) -> dict:
+    """BASELINE config #4: scale a single CR through spec updates under the
+    validating webhook (every update passes the admission chain), then
+    delete.  Returns per-transition seconds keyed ``to_<size>`` plus
+    ``delete_s`` and ``total_s``.
+
+    Spec updates ride optimistic concurrency: get-latest → mutate → update,
+    retrying Conflict (the controller updates status concurrently).
+    """
+    import time as _time
+
+    from .runtime.errors import ConflictError
+
+    mgr = stack.mgr
+    c = mgr.client
+    out = {}
+    t_start = _time.monotonic()
+
+    first, rest = sizes[0], sizes[1:]
+    req = ComposabilityRequest(
+        spec=ComposabilityRequestSpec(
+            resource=ScalarResourceDetails(
+                type="gpu", model="mi355x", size=first,
+                target_node=stack.node_name, force_detach=force_detach,
+            )
+        )
+    )
+    req.metadata.name = name
+
+    def settled(sz):
+        cur = c.try_get(ComposabilityRequest, name)
+        return (
+            cur is not None
+            and cur.status.state == "Running"
+            and len(cur.status.resources) == sz
+        )
+
+    t0 = _time.monotonic()
+    c.create(req)
+    if not mgr.wait_for(lambda: settled(first), timeout=timeout):
+        raise RuntimeError(f"churn {name}: size={first} never settled")
+    out[f"to_{first}_s"] = _time.monotonic() - t0
+
+    for sz in rest:
+        t0 = _time.monotonic()
+        for _ in range(50):
+            cur = c.get(ComposabilityRequest, name)
+            cur.spec.resource.size = sz
+            try:
+                c.update(cur)
+                break
+            except ConflictError:
+                _time.sleep(0.01)
+        else:
+            raise RuntimeError(f"churn {name}: update to size={sz} kept conflicting")
+        if not mgr.wait_for(lambda: settled(sz), timeout=timeout):
+            cur = c.try_get(ComposabilityRequest, name)
+            raise RuntimeError(
+                f"churn {name}: size={sz} never settled "
+                f"(state={cur.status.state if cur else 'gone'})"
+            )
+        out[f"to_{sz}_s"] = _time.monotonic() - t0
+
+    t0 = _time.monotonic()
+    c.delete(ComposabilityRequest, name)
+    if not mgr.wait_for(
+        lambda: c.try_get(ComposabilityRequest, name) is None, timeout=timeout
+    ):
+        raise RuntimeError(f"churn {name}: delete never completed")
+    out["delete_s"] = _time.monotonic() - t0
+    out["total_s"] = _time.monotonic() - t_start
+    return out
 
 
 def reconcile_count(stack: LocalStack) -> float:
