@@ -162,9 +162,14 @@ def main(argv=None, client: httpx.Client = None) -> int:
         return 0
 
     if args.command == "watch":
-        # k8s-style list+watch stream: synthetic ADDED replay, then live
+        # WatchList protocol (k8s 1.27+): sendInitialEvents streams ADDED
+        # per current object, an initial-events-end BOOKMARK, then live
         seen = 0
-        with http.stream("GET", f"{BASE}/{args.plural}?watch=true") as resp:
+        with http.stream(
+            "GET",
+            f"{BASE}/{args.plural}?watch=true&sendInitialEvents=true"
+            "&allowWatchBookmarks=true",
+        ) as resp:
             if resp.status_code != 200:
                 print(f"error: {resp.status_code}", file=sys.stderr)
                 return 1
@@ -172,6 +177,8 @@ def main(argv=None, client: httpx.Client = None) -> int:
                 if not line.strip():
                     continue  # keepalive
                 ev = json.loads(line)
+                if ev["type"] == "BOOKMARK":
+                    continue
                 obj = ev["object"]
                 state = (obj.get("status") or {}).get("state", "")
                 print(f'{ev["type"]:<9} {obj["metadata"]["name"]} {state}')
@@ -277,7 +284,7 @@ def main(argv=None, client: httpx.Client = None) -> int:
 
     if args.command == "delete":
         resp = http.delete(f"{BASE}/{args.plural}/{args.name}")
-        if resp.status_code != 202:
+        if resp.status_code not in (200, 202):
             return fail(resp)
         print(f"{args.plural}/{args.name} deleted")
         return 0

@@ -53,12 +53,24 @@ _PLURALS = {cls.KIND: plural for plural, cls in {
 def _raise_for(resp: httpx.Response) -> None:
     if 200 <= resp.status_code < 300:
         return
+    # error bodies are top-level metav1.Status objects (kube-apiserver wire
+    # shape); tolerate FastAPI's {"detail": ...} wrapper for foreign servers
+    reason, message = "", resp.text
     try:
-        detail = resp.json().get("detail", {})
-        reason = detail.get("reason", "") if isinstance(detail, dict) else ""
-        message = detail.get("message", str(detail)) if isinstance(detail, dict) else str(detail)
+        body = resp.json()
+        if isinstance(body, dict):
+            if body.get("kind") == "Status" or "reason" in body:
+                reason = body.get("reason", "")
+                message = body.get("message", resp.text)
+            else:
+                detail = body.get("detail", {})
+                if isinstance(detail, dict):
+                    reason = detail.get("reason", "")
+                    message = detail.get("message", str(detail))
+                else:
+                    message = str(detail)
     except ValueError:
-        reason, message = "", resp.text
+        pass
     if resp.status_code == 404 or reason == "NotFound":
         raise NotFoundError(message)
     if reason == "AlreadyExists":
@@ -186,19 +198,28 @@ class RemoteClient:
         return q
 
     def _watch_kind(self, kind: str, q: "queue.Queue[WatchEvent]") -> None:
+        """The informer protocol, exactly as client-go runs it against a
+        kube-apiserver: LIST (take ListMeta.resourceVersion) → synthesize
+        ADDED for the current objects (cache replay) → WATCH from that rv
+        with bookmarks → on 410 Expired (ERROR event) or an aged-out
+        reconnect, re-list.  Resume tokens are object resourceVersions —
+        there is no out-of-band framing field."""
         cls, plural = self._resolve(kind)
-        # the stream's initial list doubles as the informer-cache replay.
-        # Every event line carries an ``rv`` resume token; reconnects pass
-        # the last one back so the server replays only missed events from
-        # its watch-cache — an aged-out token gets an in-stream
-        # ERROR/Expired followed by a full re-list (level-triggered
-        # reconciles tolerate the replay either way).
         last_rv = -1
         while not self._stop.is_set():
             try:
-                params = {"watch": "true"}
-                if last_rv >= 0:
-                    params["resourceVersion"] = str(last_rv)
+                if last_rv < 0:
+                    resp = self._http.get(f"{BASE}/{plural}")
+                    _raise_for(resp)
+                    body = resp.json()
+                    last_rv = int(body.get("metadata", {}).get("resourceVersion", 0))
+                    for item in body["items"]:
+                        q.put(WatchEvent("ADDED", cls.model_validate(item)))
+                params = {
+                    "watch": "true",
+                    "resourceVersion": str(last_rv),
+                    "allowWatchBookmarks": "true",
+                }
                 with self._http.stream(
                     "GET", f"{BASE}/{plural}", params=params, timeout=None
                 ) as resp:
@@ -209,9 +230,18 @@ class RemoteClient:
                             continue  # keepalive
                         ev = json.loads(line)
                         if ev["type"] == "ERROR":
-                            last_rv = -1  # token expired; full replay follows
-                            continue
-                        last_rv = int(ev.get("rv", last_rv))
+                            # 410 Expired → full re-list on next loop turn
+                            last_rv = -1
+                            break
+                        obj_rv = (
+                            ev.get("object", {})
+                            .get("metadata", {})
+                            .get("resourceVersion", "")
+                        )
+                        if obj_rv:
+                            last_rv = int(obj_rv)
+                        if ev["type"] == "BOOKMARK":
+                            continue  # resume-token refresh only
                         q.put(WatchEvent(ev["type"], cls.model_validate(ev["object"])))
             except Exception as exc:
                 if self._stop.is_set():

@@ -133,8 +133,10 @@ class InMemoryStore:
             return
         self._rv = itertools.count(int(snap.get("rv", 0)) + 1)
         # seq continuity keeps pre-restart watch tokens comparable: a
-        # token < seq with an empty log → Expired → clean re-list
-        self._event_seq = int(snap.get("seq", 0))
+        # token < seq with an empty log → Expired → clean re-list.
+        # seq and rv share one counting space (apiserver parity); old
+        # snapshots carried them separately, so take the max.
+        self._event_seq = max(int(snap.get("seq", 0)), int(snap.get("rv", 0)))
         for kind, objs in snap.get("objects", {}).items():
             cls = ALL_KINDS.get(kind)
             if cls is None:
@@ -270,10 +272,14 @@ class InMemoryStore:
             ]
 
     def _notify(self, ev: WatchEvent) -> None:
+        # ONE logical resourceVersion space (apiserver/etcd-revision parity):
+        # every write bumps the object RV exactly once and the event's resume
+        # token IS that RV, so clients resume a watch from the last event
+        # object's metadata.resourceVersion exactly as k8s informers do.
         # callers hold self._lock (all mutators notify inside their
         # critical section), so the seq assignment and log append are
         # atomic with the mutation itself
-        self._event_seq += 1
+        self._event_seq = int(ev.object.metadata.resourceVersion)
         ev.seq = self._event_seq
         self._event_log.append(
             WatchEvent(ev.type, ev.object.model_copy(deep=True), None, ev.seq)
@@ -439,5 +445,9 @@ class InMemoryStore:
 
     def _finalize_delete(self, stored: K8sObject) -> K8sObject:
         del self._objects[stored.kind][stored.metadata.name]
+        # a delete is a write: it gets its own resourceVersion, and the
+        # DELETED event's object carries it (etcd-revision semantics — the
+        # resume-token space stays in lockstep with object RVs)
+        stored.metadata.resourceVersion = str(next(self._rv))
         self._notify(WatchEvent(DELETED, stored))
         return stored.model_copy(deep=True)

@@ -69,21 +69,39 @@ def _dump(obj) -> dict:
     return obj.model_dump(by_alias=True)
 
 
-def _http_error(exc: Exception) -> HTTPException:
-    """Error taxonomy over HTTP: the reason field lets a remote client map
-    back to the exact error class (409 covers both AlreadyExists and
-    optimistic-concurrency Conflict, as in the k8s API)."""
+def _status_body(code: int, reason: str, message: str) -> dict:
+    """metav1.Status — the exact error body a kube-apiserver returns, so
+    any k8s client library can consume this server's errors."""
+    return {
+        "kind": "Status",
+        "apiVersion": "v1",
+        "metadata": {},
+        "status": "Failure",
+        "message": message,
+        "reason": reason,
+        "code": code,
+    }
+
+
+def _error_response(exc: Exception):
+    """Error taxonomy over HTTP as top-level metav1.Status objects (the
+    kube-apiserver wire shape; 409 covers both AlreadyExists and
+    optimistic-concurrency Conflict, exactly as in the k8s API)."""
+    from fastapi.responses import JSONResponse
+
     if isinstance(exc, NotFoundError):
-        return HTTPException(404, {"reason": "NotFound", "message": str(exc)})
-    if isinstance(exc, AlreadyExistsError):
-        return HTTPException(409, {"reason": "AlreadyExists", "message": str(exc)})
-    if isinstance(exc, ConflictError):
-        return HTTPException(409, {"reason": "Conflict", "message": str(exc)})
-    if isinstance(exc, AdmissionDenied):
-        return HTTPException(403, {"reason": "Forbidden", "message": str(exc)})
-    if isinstance(exc, ValueError):
-        return HTTPException(422, {"reason": "Invalid", "message": str(exc)})
-    return HTTPException(500, {"reason": "InternalError", "message": str(exc)})
+        body = _status_body(404, "NotFound", str(exc))
+    elif isinstance(exc, AlreadyExistsError):
+        body = _status_body(409, "AlreadyExists", str(exc))
+    elif isinstance(exc, ConflictError):
+        body = _status_body(409, "Conflict", str(exc))
+    elif isinstance(exc, AdmissionDenied):
+        body = _status_body(403, "Forbidden", str(exc))
+    elif isinstance(exc, ValueError):
+        body = _status_body(422, "Invalid", str(exc))
+    else:
+        body = _status_body(500, "InternalError", str(exc))
+    return JSONResponse(status_code=body["code"], content=body)
 
 
 def build_app(client: Client, token: str = None) -> FastAPI:
@@ -153,38 +171,86 @@ def build_app(client: Client, token: str = None) -> FastAPI:
             media_type=prometheus_client.CONTENT_TYPE_LATEST,
         )
 
-    @app.get(BASE + "/{plural}")
+    def _current_rv() -> int:
+        store = getattr(client, "store", None)
+        return store.current_seq() if store is not None else 0
+
     async def list_objects(
         plural: str,
         labelSelector: str = "",
         watch: bool = False,
-        resourceVersion: int = -1,
+        resourceVersion: str = "",
+        allowWatchBookmarks: bool = False,
+        sendInitialEvents: bool = False,
+        timeoutSeconds: float = 0,
     ):
         cls = _cls(plural)
         if watch:
-            return await _watch_stream(cls, resourceVersion)
+            return await _watch_stream(
+                cls, resourceVersion, allowWatchBookmarks,
+                sendInitialEvents, timeoutSeconds,
+            )
         labels = None
         if labelSelector:
             labels = dict(part.split("=", 1) for part in labelSelector.split(","))
         items = client.list(cls, labels)
+        # ListMeta.resourceVersion is the watch-resume token for the
+        # list-then-watch informer protocol (apiserver parity)
         return {
-            "apiVersion": API_VERSION,
+            "apiVersion": cls.model_fields["apiVersion"].default,
             "kind": cls.KIND + "List",
+            "metadata": {"resourceVersion": str(_current_rv())},
             "items": [_dump(o) for o in items],
         }
 
-    async def _watch_stream(cls, resource_version: int = -1):
-        """k8s-style list+watch, newline-delimited JSON. Every event line
-        carries ``rv`` (the store's event sequence) as a resume token:
-        reconnecting with ``?resourceVersion=<rv>`` replays only missed
-        events from the bounded watch-cache; an aged-out token gets one
-        ``{"type": "ERROR", "reason": "Expired"}`` line followed by the
-        full list replay (the apiserver's 410-Gone contract, in-stream)."""
+    async def _watch_stream(
+        cls,
+        resource_version: str = "",
+        bookmarks: bool = False,
+        send_initial: bool = False,
+        timeout_seconds: float = 0,
+    ):
+        """kube-apiserver watch semantics, one JSON WatchEvent per line:
+
+        * ``?resourceVersion=N``        → stream events after N; an aged-out
+          N gets ONE ``{"type":"ERROR","object":metav1.Status(410 Expired)}``
+          line and the stream ends — the client re-lists (the real 410-Gone
+          contract; no in-stream re-list).
+        * no rv / ``resourceVersion=0`` → stream from "now" (clients that
+          want state do list-then-watch, as informers do).
+        * ``sendInitialEvents=true``    → the 1.27+ WatchList protocol:
+          synthetic ADDED per current object, then a BOOKMARK annotated
+          ``k8s.io/initial-events-end`` at the snapshot rv, then live events.
+        * ``allowWatchBookmarks=true``  → periodic BOOKMARK events carry the
+          current rv so idle clients keep a fresh resume token.
+        * every event object's ``metadata.resourceVersion`` IS the resume
+          token (object RVs and watch tokens share one counting space).
+        """
         import asyncio
         import json as _json
         import queue as _queue
 
         from fastapi.responses import StreamingResponse
+
+        # absent rv → watch from "now" (k8s watch default). A PRESENT rv —
+        # including "0", which a fresh store's list legitimately returns —
+        # resumes from the event log, so a client that lists an empty store
+        # and watches from its rv never loses events created in the gap
+        # (k8s "0" means "any version"; exact replay satisfies it).
+        try:
+            rv = int(resource_version) if resource_version != "" else None
+        except ValueError:
+            rv = None
+
+        def bookmark_line(at_rv: int, initial_end: bool = False) -> str:
+            obj = {
+                "kind": cls.KIND,
+                "apiVersion": cls.model_fields["apiVersion"].default,
+                "metadata": {"resourceVersion": str(at_rv)},
+            }
+            if initial_end:
+                obj["metadata"]["annotations"] = {"k8s.io/initial-events-end": "true"}
+            return _json.dumps({"type": "BOOKMARK", "object": obj}) + "\n"
 
         async def gen():
             import os
@@ -192,32 +258,40 @@ def build_app(client: Client, token: str = None) -> FastAPI:
 
             # k8s watch-timeout contract: every stream is closed server-side
             # after a bounded lifetime and the client reconnects (cheap with
-            # rv resume tokens). Bounds the damage of half-open connections
-            # — a client reading keepalives from a stale server otherwise
-            # never notices it should reconnect.
-            lifetime = float(os.environ.get("CRO_WATCH_TIMEOUT", "300"))
+            # rv resume tokens). Bounds the damage of half-open connections.
+            lifetime = timeout_seconds or float(os.environ.get("CRO_WATCH_TIMEOUT", "300"))
             stream_deadline = _time.monotonic() + lifetime
             events = client.watch([cls.KIND])
             store = getattr(client, "store", None)
             last_seq = 0
             try:
-                buffered = None
-                if resource_version >= 0 and store is not None:
-                    buffered = store.events_since(resource_version, [cls.KIND])
-                if buffered is not None:
+                if send_initial:
+                    snapshot_rv = _current_rv()
+                    for obj in client.list(cls):
+                        yield _json.dumps(
+                            {"type": "ADDED", "object": _dump(obj)}
+                        ) + "\n"
+                    yield bookmark_line(snapshot_rv, initial_end=True)
+                    last_seq = snapshot_rv
+                elif rv is not None and store is not None:
+                    buffered = store.events_since(rv, [cls.KIND])
+                    if buffered is None:
+                        # aged-out token → 410 Expired, stream ends
+                        yield _json.dumps({
+                            "type": "ERROR",
+                            "object": _status_body(
+                                410, "Expired",
+                                f"too old resource version: {rv}"),
+                        }) + "\n"
+                        return
                     for ev in buffered:
                         last_seq = ev.seq
                         yield _json.dumps(
-                            {"type": ev.type, "object": _dump(ev.object), "rv": ev.seq}
+                            {"type": ev.type, "object": _dump(ev.object)}
                         ) + "\n"
+                    last_seq = max(last_seq, rv)
                 else:
-                    if resource_version >= 0:
-                        yield _json.dumps({"type": "ERROR", "reason": "Expired"}) + "\n"
-                    snapshot_rv = store.current_seq() if store is not None else 0
-                    for obj in client.list(cls):
-                        yield _json.dumps(
-                            {"type": "ADDED", "object": _dump(obj), "rv": snapshot_rv}
-                        ) + "\n"
+                    last_seq = _current_rv()  # watch from "now"
                 loop = asyncio.get_running_loop()
                 while True:
                     if _time.monotonic() >= stream_deadline:
@@ -225,7 +299,10 @@ def build_app(client: Client, token: str = None) -> FastAPI:
                     try:
                         ev = await loop.run_in_executor(None, events.get, True, 1.0)
                     except _queue.Empty:
-                        yield "\n"  # keepalive; also surfaces disconnects
+                        if bookmarks:
+                            yield bookmark_line(max(last_seq, _current_rv()))
+                        else:
+                            yield "\n"  # legacy keepalive
                         continue
                     except RuntimeError:
                         return  # event loop / executor shutting down
@@ -233,7 +310,7 @@ def build_app(client: Client, token: str = None) -> FastAPI:
                         continue  # already served from the resume buffer
                     last_seq = ev.seq or last_seq
                     yield _json.dumps(
-                        {"type": ev.type, "object": _dump(ev.object), "rv": ev.seq}
+                        {"type": ev.type, "object": _dump(ev.object)}
                     ) + "\n"
             finally:
                 # disconnects must release the watcher or every later event
@@ -242,26 +319,23 @@ def build_app(client: Client, token: str = None) -> FastAPI:
                 if store is not None and hasattr(store, "stop_watch"):
                     store.stop_watch(events)
 
-        return StreamingResponse(gen(), media_type="application/x-ndjson")
+        return StreamingResponse(gen(), media_type="application/json")
 
-    @app.get(BASE + "/{plural}/{name:path}")
     def get_object(plural: str, name: str):
         cls = _cls(plural)
         try:
             return _dump(client.get(cls, name))
         except Exception as exc:
-            raise _http_error(exc)
+            return _error_response(exc)
 
-    @app.post(BASE + "/{plural}", status_code=201)
     async def create_object(plural: str, request: Request):
         cls = _cls(plural)
         try:
             obj = cls.model_validate(await request.json())
             return _dump(client.create(obj))
         except Exception as exc:
-            raise _http_error(exc)
+            return _error_response(exc)
 
-    @app.put(BASE + "/{plural}/{name:path}")
     async def update_object(plural: str, name: str, request: Request):
         cls = _cls(plural)
         if name.endswith("/status"):
@@ -271,21 +345,41 @@ def build_app(client: Client, token: str = None) -> FastAPI:
                 obj.metadata.name = name
                 return _dump(client.update_status(obj))
             except Exception as exc:
-                raise _http_error(exc)
+                return _error_response(exc)
         try:
             obj = cls.model_validate(await request.json())
             obj.metadata.name = name
             return _dump(client.update(obj))
         except Exception as exc:
-            raise _http_error(exc)
+            return _error_response(exc)
 
-    @app.delete(BASE + "/{plural}/{name:path}", status_code=202)
     def delete_object(plural: str, name: str):
         cls = _cls(plural)
         try:
             client.delete(cls, name)
         except Exception as exc:
-            raise _http_error(exc)
-        return {"status": "deleted"}
+            return _error_response(exc)
+        # metav1.Status success body, as the apiserver returns for deletes
+        return {
+            "kind": "Status", "apiVersion": "v1", "metadata": {},
+            "status": "Success",
+            "details": {"name": name, "kind": plural},
+        }
+
+    # Register the handlers under every API group the kinds belong to:
+    # the home CRD group plus the REAL k8s groups for the native kinds,
+    # so k8s tooling addresses ResourceSlices/DeviceTaintRules/Leases at
+    # their canonical paths (resource.k8s.io is what the reference writes,
+    # internal/utils/gpus.go:894-989).
+    for base in (
+        BASE,
+        "/apis/resource.k8s.io/v1alpha3",
+        "/apis/coordination.k8s.io/v1",
+    ):
+        app.get(base + "/{plural}")(list_objects)
+        app.get(base + "/{plural}/{name:path}")(get_object)
+        app.post(base + "/{plural}", status_code=201)(create_object)
+        app.put(base + "/{plural}/{name:path}")(update_object)
+        app.delete(base + "/{plural}/{name:path}")(delete_object)
 
     return app

@@ -40,7 +40,9 @@ def test_create_get_list_delete(api_stack):
         timeout=10,
     )
 
-    assert http.delete(f"{PLURAL}/r1").status_code == 202
+    resp = http.delete(f"{PLURAL}/r1")
+    assert resp.status_code == 200  # metav1.Status success body
+    assert resp.json()["kind"] == "Status" and resp.json()["status"] == "Success"
     assert stack.mgr.wait_for(
         lambda: http.get(f"{PLURAL}/r1").status_code == 404, timeout=10
     )
@@ -51,7 +53,8 @@ def test_admission_enforced_over_http(api_stack):
     bad = make_request("r1", policy="differentnode", target_node="node0")
     resp = http.post(PLURAL, json=bad.model_dump(by_alias=True))
     assert resp.status_code == 403
-    assert "TargetNode cannot be specified" in resp.json()["detail"]["message"]
+    assert resp.json()["kind"] == "Status"
+    assert "TargetNode cannot be specified" in resp.json()["message"]
 
 
 def test_schema_validation_over_http(api_stack):

@@ -271,9 +271,10 @@ def test_remote_client_credentials(monkeypatch):
 
 
 def test_watch_resume_token(api_server):
-    """Reconnecting with ?resourceVersion=<rv> replays only missed events
-    from the server watch-cache; an aged-out token gets ERROR/Expired then
-    a full list replay (in-stream 410-Gone contract)."""
+    """kube-apiserver watch semantics: list → take ListMeta.resourceVersion
+    → watch from it (only missed events replay); an aged-out token gets ONE
+    ERROR event carrying a metav1.Status 410 Expired and the stream ENDS
+    (the client re-lists — the real 410-Gone contract)."""
     import json
 
     import httpx
@@ -283,15 +284,10 @@ def test_watch_resume_token(api_server):
     server_mgr.client.create(make_request("b", model="mi300x", target_node="node0"))
 
     base = f"{url}/apis/cro.hpsys.ibm.ie.com/v1alpha1/composabilityrequests"
-    with httpx.stream("GET", base, params={"watch": "true"}, timeout=10) as resp:
-        lines = []
-        for line in resp.iter_lines():
-            if line.strip():
-                lines.append(json.loads(line))
-            if len(lines) == 2:
-                break
-    assert [l["object"]["metadata"]["name"] for l in lines] == ["a", "b"]
-    rv = lines[-1]["rv"]
+    listed = httpx.get(base, timeout=10).json()
+    assert sorted(i["metadata"]["name"] for i in listed["items"]) == ["a", "b"]
+    rv = int(listed["metadata"]["resourceVersion"])
+    assert rv > 0
 
     # miss one event while disconnected
     server_mgr.client.create(
@@ -307,23 +303,54 @@ def test_watch_resume_token(api_server):
                 break
     assert first["type"] == "ADDED"
     assert first["object"]["metadata"]["name"] == "c"  # ONLY the missed event
-    assert first["rv"] > rv
+    # the resume token IS the object resourceVersion — no framing field
+    assert "rv" not in first
+    assert int(first["object"]["metadata"]["resourceVersion"]) > rv
 
-    # aged-out token → ERROR then full replay
+    # aged-out token → ONE 410 Expired Status, stream ends (no replay)
     store = server_mgr.store
     while store._event_log:
         store._event_log.popleft()
     with httpx.stream(
         "GET", base, params={"watch": "true", "resourceVersion": "1"}, timeout=10
     ) as resp:
+        got = [json.loads(l) for l in resp.iter_lines() if l.strip()]
+    assert len(got) == 1
+    err = got[0]
+    assert err["type"] == "ERROR"
+    assert err["object"]["kind"] == "Status"
+    assert err["object"]["code"] == 410
+    assert err["object"]["reason"] == "Expired"
+
+
+def test_watch_sendinitialevents(api_server):
+    """The 1.27+ WatchList protocol: sendInitialEvents=true streams ADDED
+    per current object then a BOOKMARK annotated k8s.io/initial-events-end
+    at the snapshot rv."""
+    import json
+
+    import httpx
+
+    url, server_mgr = api_server
+    server_mgr.client.create(make_request("w1", target_node="node0"))
+    server_mgr.client.create(make_request("w2", model="mi300x", target_node="node0"))
+    base = f"{url}/apis/cro.hpsys.ibm.ie.com/v1alpha1/composabilityrequests"
+    with httpx.stream(
+        "GET", base,
+        params={"watch": "true", "sendInitialEvents": "true",
+                "allowWatchBookmarks": "true"},
+        timeout=10,
+    ) as resp:
         got = []
         for line in resp.iter_lines():
             if line.strip():
                 got.append(json.loads(line))
-            if len(got) == 4:
+            if len(got) == 3:
                 break
-    assert got[0] == {"type": "ERROR", "reason": "Expired"}
-    assert sorted(l["object"]["metadata"]["name"] for l in got[1:]) == ["a", "b", "c"]
+    assert [g["type"] for g in got] == ["ADDED", "ADDED", "BOOKMARK"]
+    bm = got[2]["object"]
+    assert bm["metadata"]["annotations"] == {"k8s.io/initial-events-end": "true"}
+    assert int(bm["metadata"]["resourceVersion"]) > 0
 
 
 def test_store_events_since_semantics():
