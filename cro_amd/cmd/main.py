@@ -72,6 +72,14 @@ def main(argv=None) -> int:
     p.add_argument("--webhook-insecure", action="store_true",
                    help="serve the webhook over plain HTTP (tests only; "
                    "a real apiserver requires TLS)")
+    p.add_argument("--metrics-cert-dir",
+                   default="/tmp/k8s-metrics-server/serving-certs",
+                   help="directory with tls.crt/tls.key for the metrics "
+                   "listener (cert-manager mount); without certs the "
+                   "metrics port serves plain HTTP with a warning")
+    p.add_argument("--metrics-secure", action="store_true", default=True,
+                   help="require the bearer token on the dedicated metrics "
+                   "port (reference --metrics-secure parity; default on)")
     p.add_argument("--node", default=os.environ.get("NODE_NAME", ""),
                    help="local node name this operator instance manages")
     p.add_argument("--api-server", default="",
@@ -130,21 +138,27 @@ def main(argv=None) -> int:
     token_dir = args.data_dir or "/var/run/cro-amd"
     api_token = os.environ.get("CRO_API_TOKEN", "")
     agent_token = os.environ.get("CRO_AGENT_TOKEN", "")
+    metrics_token = os.environ.get("CRO_METRICS_TOKEN", "")
     generated = {}
     if not api_token:
         import secrets
 
         api_token = secrets.token_hex(24)
-        generated["api.token"] = api_token
+        generated["api.token"] = ("CRO_API_TOKEN", api_token)
     if not agent_token and args.node:
         import secrets
 
         agent_token = secrets.token_hex(24)
-        generated["agent.token"] = agent_token
+        generated["agent.token"] = ("CRO_AGENT_TOKEN", agent_token)
+    if not metrics_token and args.metrics_secure:
+        import secrets
+
+        metrics_token = secrets.token_hex(24)
+        generated["metrics.token"] = ("CRO_METRICS_TOKEN", metrics_token)
     if generated:
         try:
             os.makedirs(token_dir, exist_ok=True)
-            for fname, value in generated.items():
+            for fname, (envname, value) in generated.items():
                 path = os.path.join(token_dir, fname)
                 fd = os.open(path, os.O_WRONLY | os.O_CREAT | os.O_TRUNC, 0o600)
                 with os.fdopen(fd, "w") as f:
@@ -152,12 +166,14 @@ def main(argv=None) -> int:
                 log.warning(
                     "no %s configured; generated one at %s (set the env var "
                     "from a Secret for multi-process deployments)",
-                    "CRO_API_TOKEN" if fname == "api.token" else "CRO_AGENT_TOKEN",
-                    path,
+                    envname, path,
                 )
         except OSError as exc:
             log.warning("could not persist generated tokens (%s); clients "
                         "of this process must read them from the log", exc)
+    # the shared API app's /metrics route reads CRO_METRICS_TOKEN directly
+    if metrics_token:
+        os.environ["CRO_METRICS_TOKEN"] = metrics_token
 
     from ..controllers import build_manager
     from ..fabric.adapter import new_adapter
@@ -187,7 +203,6 @@ def main(argv=None) -> int:
         max_concurrent_reconciles=args.max_concurrent_reconciles,
         enable_webhook=os.environ.get("ENABLE_WEBHOOKS", "") != "false",
         syncer_period=args.syncer_period,
-        metrics_port=parse_port(args.metrics_bind_address, 8443),
     )
     # FTI/NEC providers resolve Node→machine through the cluster client,
     # which does not exist until the manager is built — wire it now (a
@@ -363,6 +378,38 @@ def main(argv=None) -> int:
                 args.webhook_cert_dir,
             )
 
+    # -- dedicated authenticated metrics listener ---------------------------
+    # The reference serves :8443 metrics over HTTPS behind authn/authz
+    # (cmd/main.go:109-127); here: TLS from --metrics-cert-dir + the bearer
+    # token the ServiceMonitor sends from the cro-amd-tokens Secret.
+    metrics_server = None
+    if args.metrics_bind_address:
+        from ..server.metrics import MetricsServer
+
+        metrics_server = MetricsServer(
+            parse_port(args.metrics_bind_address, 8443),
+            token=metrics_token if args.metrics_secure else "",
+            cert_dir=args.metrics_cert_dir,
+            certfile=args.tls_cert_file or None,
+            keyfile=args.tls_key_file or None,
+        )
+        metrics_server.start()
+        log.info(
+            "metrics serving on %s (%s, %s)",
+            args.metrics_bind_address,
+            "TLS" if metrics_server.tls else "plain HTTP",
+            "bearer-token" if (metrics_token and args.metrics_secure) else "open",
+        )
+
+    # -- kubelet health-probe listener (manager.yaml probes on :8081) -------
+    health_server = None
+    if args.health_probe_bind_address:
+        from ..server.metrics import HealthServer
+
+        health_server = HealthServer(
+            parse_port(args.health_probe_bind_address, 8081))
+        health_server.start()
+
     # -- leader election + controller start ---------------------------------
     elector = None
     if not args.serve_only:
@@ -406,6 +453,10 @@ def main(argv=None) -> int:
         webhook_server.should_exit = True
         if webhook_thread is not None:
             webhook_thread.join(timeout=5)
+    if metrics_server is not None:
+        metrics_server.stop()
+    if health_server is not None:
+        health_server.stop()
     mgr.stop()
     if store is not None:
         store.close()  # final flush of the durable snapshot
