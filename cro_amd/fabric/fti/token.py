@@ -51,6 +51,46 @@ def file_credentials(path: str) -> CredentialsFn:
     return load
 
 
+SECRET_KEYS = ("username", "password", "client_id", "client_secret", "realm")
+
+
+def secret_dir_credentials(path: str) -> CredentialsFn:
+    """Kubernetes mounted-Secret credential source (token.go:103-127
+    parity): the RBAC-pinned ``credentials`` Secret mounts as a directory
+    of per-key files (username/password/client_id/client_secret/realm).
+    Every load re-reads the files, so a rotated Secret is picked up at the
+    next token refresh without a restart (kubelet atomically updates the
+    mount's symlinks)."""
+
+    def load() -> Dict[str, str]:
+        creds: Dict[str, str] = {}
+        for key in SECRET_KEYS:
+            try:
+                with open(os.path.join(path, key)) as f:
+                    creds[key] = f.read().strip()
+            except OSError:
+                creds[key] = ""
+        return creds
+
+    return load
+
+
+def default_credentials() -> CredentialsFn:
+    """Production credential resolution, checked per token fetch:
+
+    1. ``CRO_FTI_CREDENTIALS_DIR``  — mounted k8s Secret (rotation-aware);
+    2. ``CRO_FTI_CREDENTIALS_FILE`` — JSON file;
+    3. ``CRO_FTI_*`` env variables.
+    """
+    cred_dir = os.environ.get("CRO_FTI_CREDENTIALS_DIR", "")
+    if cred_dir:
+        return secret_dir_credentials(cred_dir)
+    cred_file = os.environ.get("CRO_FTI_CREDENTIALS_FILE", "")
+    if cred_file:
+        return file_credentials(cred_file)
+    return env_credentials
+
+
 def parse_jwt_expiry(access_token: str) -> float:
     parts = access_token.split(".")
     if len(parts) != 3:
@@ -73,7 +113,7 @@ class CachedToken:
         if not endpoint.endswith("/"):
             endpoint += "/"
         self.endpoint = endpoint
-        self.credentials = credentials or env_credentials
+        self.credentials = credentials or default_credentials()
         self.transport = transport
         self.verify = verify
         self.leeway = EXPIRY_LEEWAY

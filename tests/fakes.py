@@ -143,6 +143,11 @@ class FakeFTIServer:
 
         if "id_manager" in path and path.endswith("/token"):
             self.token_calls += 1
+            from urllib.parse import parse_qsl
+
+            self.last_token_request = dict(
+                parse_qsl(request.content.decode(errors="replace"))
+            )
             if self.token_persona == "bad-creds":
                 return httpx.Response(401, json={"error": "invalid_grant"})
             if self.token_persona == "non-json":

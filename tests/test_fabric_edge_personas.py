@@ -135,3 +135,65 @@ def test_token_non_json_error_body():
     tok = CachedToken("f.example", credentials=CREDS, transport=httpx.MockTransport(handler))
     with pytest.raises(TokenError, match="500"):
         tok.get_token()
+
+
+def test_secret_dir_credentials(tmp_path):
+    """Mounted-Secret credential source: per-key files, re-read per load
+    (token.go:103-127 parity)."""
+    from cro_amd.fabric.fti.token import secret_dir_credentials
+
+    for key, val in [("username", "u"), ("password", "p"),
+                     ("client_id", "cid"), ("client_secret", "cs"),
+                     ("realm", "r1")]:
+        (tmp_path / key).write_text(val + "\n")
+    creds = secret_dir_credentials(str(tmp_path))
+    assert creds() == {"username": "u", "password": "p", "client_id": "cid",
+                       "client_secret": "cs", "realm": "r1"}
+    # missing key → empty string, not a crash
+    (tmp_path / "realm").unlink()
+    assert creds()["realm"] == ""
+
+
+def test_secret_rotation_picked_up_on_refresh(tmp_path):
+    """A rotated Secret is used at the NEXT token refresh without restart:
+    the credential fn re-reads the mount per fetch."""
+    import time
+
+    from cro_amd.fabric.fti.token import CachedToken, secret_dir_credentials
+    from tests.fakes import FakeFTIServer
+
+    for key, val in [("username", "old-user"), ("password", "p"),
+                     ("client_id", "c"), ("client_secret", "s"),
+                     ("realm", "r")]:
+        (tmp_path / key).write_text(val)
+
+    server = FakeFTIServer()
+    server.token_exp = time.time() + 10  # inside leeway → refetch each call
+    tok = CachedToken("fabric.example",
+                      credentials=secret_dir_credentials(str(tmp_path)),
+                      transport=server.transport())
+    tok.get_token()
+    assert server.last_token_request["username"] == "old-user"
+
+    (tmp_path / "username").write_text("new-user")  # kubelet rotates the Secret
+    tok.get_token()
+    assert server.last_token_request["username"] == "new-user"
+
+
+def test_default_credentials_resolution(tmp_path, monkeypatch):
+    from cro_amd.fabric.fti.token import default_credentials
+
+    monkeypatch.delenv("CRO_FTI_CREDENTIALS_DIR", raising=False)
+    monkeypatch.delenv("CRO_FTI_CREDENTIALS_FILE", raising=False)
+    monkeypatch.setenv("CRO_FTI_USERNAME", "env-user")
+    assert default_credentials()()["username"] == "env-user"
+
+    (tmp_path / "creds.json").write_text('{"username": "file-user"}')
+    monkeypatch.setenv("CRO_FTI_CREDENTIALS_FILE", str(tmp_path / "creds.json"))
+    assert default_credentials()()["username"] == "file-user"
+
+    d = tmp_path / "secret"
+    d.mkdir()
+    (d / "username").write_text("dir-user")
+    monkeypatch.setenv("CRO_FTI_CREDENTIALS_DIR", str(d))
+    assert default_credentials()()["username"] == "dir-user"  # dir wins
