@@ -76,19 +76,27 @@ def secret_dir_credentials(path: str) -> CredentialsFn:
 
 
 def default_credentials() -> CredentialsFn:
-    """Production credential resolution, checked per token fetch:
+    """Production credential resolution, re-checked per token fetch:
 
     1. ``CRO_FTI_CREDENTIALS_DIR``  — mounted k8s Secret (rotation-aware);
+       skipped while the optional mount is absent/empty, so the manifest
+       can set the path unconditionally;
     2. ``CRO_FTI_CREDENTIALS_FILE`` — JSON file;
     3. ``CRO_FTI_*`` env variables.
     """
-    cred_dir = os.environ.get("CRO_FTI_CREDENTIALS_DIR", "")
-    if cred_dir:
-        return secret_dir_credentials(cred_dir)
-    cred_file = os.environ.get("CRO_FTI_CREDENTIALS_FILE", "")
-    if cred_file:
-        return file_credentials(cred_file)
-    return env_credentials
+
+    def load() -> Dict[str, str]:
+        cred_dir = os.environ.get("CRO_FTI_CREDENTIALS_DIR", "")
+        if cred_dir and any(
+            os.path.exists(os.path.join(cred_dir, k)) for k in SECRET_KEYS
+        ):
+            return secret_dir_credentials(cred_dir)()
+        cred_file = os.environ.get("CRO_FTI_CREDENTIALS_FILE", "")
+        if cred_file and os.path.exists(cred_file):
+            return file_credentials(cred_file)()
+        return env_credentials()
+
+    return load
 
 
 def parse_jwt_expiry(access_token: str) -> float:
