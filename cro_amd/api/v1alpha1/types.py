@@ -261,6 +261,9 @@ class BareMetalHost(K8sObject):
 class DeviceConfigDriver(_Model):
     enable: bool = False
     daemonset_name: str = "amd-gpu-driver"
+    # rootfs of the driver container on the host — module ops chroot here
+    # in container mode (the /run/nvidia/driver analog, gpus.go:566-749)
+    driver_root: str = "/run/amdgpu-driver"
 
 
 class DeviceConfigSpec(_Model):

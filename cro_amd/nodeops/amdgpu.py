@@ -153,6 +153,42 @@ class AmdNodeOps(NodeOps):
             return "host"
         return "none"
 
+    def driver_root(self, node: str) -> str:
+        """Driver-container rootfs for module operations, or "" in host
+        mode.  Container mode resolves DeviceConfig.spec.driver.driver_root
+        (CRO_DRIVER_ROOT env overrides) — the chroot target of the
+        reference's drain path B (gpus.go:566-749: chroot
+        /run/nvidia/driver before module/device ops)."""
+        if self.driver_mode(node) != "container":
+            return ""
+        import os
+
+        env_root = os.environ.get("CRO_DRIVER_ROOT", "")
+        if env_root:
+            return env_root
+        from ..api.v1alpha1.types import DeviceConfig
+
+        for dc in self.client.list(DeviceConfig):
+            if dc.spec.driver.enable:
+                return dc.spec.driver.driver_root
+        return "/run/amdgpu-driver"
+
+    def _module_argv(self, node: str, argv: List[str]) -> List[str]:
+        """Wrap a module-level command (modprobe, croagent probe) with the
+        driver-container chroot when the driver is containerized — the
+        kernel modules and ROCm userspace live in the driver container's
+        rootfs there, not on the host."""
+        root = self.driver_root(node)
+        if root:
+            return ["chroot", root] + argv
+        return argv
+
+    def probe_argv_prefix(self, node: str) -> List[str]:
+        """Chroot prefix for the exec-probe hook (probe.make_exec_probe_fn)
+        — [] in host mode, ["chroot", <driver_root>] in container mode."""
+        root = self.driver_root(node)
+        return ["chroot", root] if root else []
+
     def ensure_driver(self, node: str) -> None:
         """Driver gate before any attach work.
 
@@ -376,9 +412,13 @@ class AmdNodeOps(NodeOps):
         # time while KFD tears down, and the remove can stall the writer —
         # run asynchronously and report progress on re-checks, the pattern
         # the reference uses for its async sysfs remove (gpus.go:1534-1585)
+        # resolved before the thread starts: the chroot decision must not
+        # race a DeviceConfig change mid-drain
+        modprobe_argv = self._module_argv(node, ["modprobe", "-r", "amdgpu"])
+
         def unload_and_remove():
             try:
-                rc, _, err = self.execer.run(node, ["modprobe", "-r", "amdgpu"], timeout=120)
+                rc, _, err = self.execer.run(node, modprobe_argv, timeout=120)
                 if rc != 0:
                     raise ExecError(f"modprobe -r amdgpu failed: {err}", rc=rc, stderr=err)
                 self.execer.write_file(

@@ -136,16 +136,20 @@ def probe_fn_for_nodeops(gpu) -> dict:
     return run_probe(dev)
 
 
-def probe_via_exec(execer, node: str, gpu) -> dict:
+def probe_via_exec(execer, node: str, gpu, argv_prefix=None) -> dict:
     """Health probe through the node agent (``croagent probe --bdf``) — the
     cluster shape where controllers run off-node and reach hardware only
     through the NodeExec seam.  Returns the same dict shape as run_probe.
+
+    ``argv_prefix`` wraps the command for the container-driver arm (e.g.
+    ``["chroot", "/run/amdgpu-driver"]`` so the probe runs against the
+    driver container's ROCm userspace — the nvidia-smi-through-chroot
+    analog, gpus.go:566-749).
     """
     import json as _json
 
-    rc, out, err = execer.run(
-        node, ["croagent", "probe", "--bdf", gpu.pci_bdf], timeout=300
-    )
+    argv = list(argv_prefix or []) + ["croagent", "probe", "--bdf", gpu.pci_bdf]
+    rc, out, err = execer.run(node, argv, timeout=300)
     if rc != 0 and not out.strip():
         return {"ok": False, "rc": rc, "msg": err.strip() or "croagent probe failed"}
     try:
@@ -154,10 +158,16 @@ def probe_via_exec(execer, node: str, gpu) -> dict:
         return {"ok": False, "rc": rc, "msg": f"unparseable probe output: {out[:200]}"}
 
 
-def make_exec_probe_fn(execer, node: str):
-    """AmdNodeOps probe hook bound to a NodeExec (local or remote agent)."""
+def make_exec_probe_fn(execer, node: str, argv_prefix_fn=None):
+    """AmdNodeOps probe hook bound to a NodeExec (local or remote agent).
+
+    ``argv_prefix_fn()`` is resolved per probe so a driver-mode change
+    (DeviceConfig applied mid-flight) switches the chroot arm without a
+    rebuild — pass ``lambda: ops.probe_argv_prefix(node)``.
+    """
 
     def probe(gpu):
-        return probe_via_exec(execer, node, gpu)
+        prefix = argv_prefix_fn() if argv_prefix_fn else None
+        return probe_via_exec(execer, node, gpu, argv_prefix=prefix)
 
     return probe

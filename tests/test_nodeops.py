@@ -539,3 +539,105 @@ def test_driver_detection_chain():
     # none: fresh ops without module or DeviceConfig
     bare = AmdNodeOps(MockNodeExec(), cdi_dir="/etc/cdi")
     assert bare.driver_mode(NODE) == "none"
+
+
+def _container_driver_client(driver_root="/run/amdgpu-driver"):
+    from cro_amd.api.v1alpha1.types import DaemonSet, DeviceConfig
+    from cro_amd.runtime.client import Client
+    from cro_amd.runtime.store import InMemoryStore
+
+    client = Client(InMemoryStore())
+    dc = DeviceConfig()
+    dc.metadata.name = "default"
+    dc.spec.driver.enable = True
+    dc.spec.driver.driver_root = driver_root
+    client.create(dc)
+    ds = DaemonSet()
+    ds.metadata.name = "amd-gpu-operator/amd-gpu-driver"
+    ds.status.desired_number_scheduled = 1
+    ds.status.number_ready = 1
+    client.create(ds)
+    return client
+
+
+def test_last_gpu_drain_container_arm_chroots_modprobe(monkeypatch):
+    """Drain path B (gpus.go:566-749 parity): with a containerized driver
+    the module unload runs through the driver container's rootfs —
+    ``chroot <driver_root> modprobe -r amdgpu`` — because the kernel
+    modules live there, not on the host."""
+    import time
+
+    from cro_amd.nodeops.amdgpu import DrainInProgress
+
+    monkeypatch.delenv("CRO_DRIVER_ROOT", raising=False)
+    ex = MockNodeExec()
+    ids = kfd_fixture(ex, 1)
+    ex.set_command(("chroot", "/run/amdgpu-driver", "modprobe", "-r", "amdgpu"), (0, "", ""))
+    ops = AmdNodeOps(ex, client=_container_driver_client(), destructive=True)
+    assert ops.driver_mode(NODE) == "container"
+    assert ops.driver_root(NODE) == "/run/amdgpu-driver"
+
+    with pytest.raises(DrainInProgress):
+        ops.drain(NODE, ids[0])
+    time.sleep(0.05)
+    ops.drain(NODE, ids[0])  # completion check
+    assert ("run", NODE, ("chroot", "/run/amdgpu-driver", "modprobe", "-r", "amdgpu")) in ex.calls
+    # bare modprobe must NOT have run on the host
+    assert ("run", NODE, ("modprobe", "-r", "amdgpu")) not in ex.calls
+    # sysfs hot-remove stays host-global (sysfs is shared with the container)
+    assert ex.files[(NODE, "/sys/bus/pci/devices/0000:03:00.0/remove")] == "1"
+
+
+def test_last_gpu_drain_host_arm_plain_modprobe(monkeypatch):
+    """Host-driver arm: no chroot wrapping (drain path A)."""
+    import time
+
+    from cro_amd.nodeops.amdgpu import DrainInProgress
+
+    monkeypatch.delenv("CRO_DRIVER_ROOT", raising=False)
+    ex = MockNodeExec()
+    ids = kfd_fixture(ex, 1)
+    ex.set_command(("modprobe", "-r", "amdgpu"), (0, "", ""))
+    ops = AmdNodeOps(ex, destructive=True)
+    assert ops.driver_mode(NODE) == "host"
+    assert ops.driver_root(NODE) == ""
+    with pytest.raises(DrainInProgress):
+        ops.drain(NODE, ids[0])
+    time.sleep(0.05)
+    ops.drain(NODE, ids[0])
+    assert ("run", NODE, ("modprobe", "-r", "amdgpu")) in ex.calls
+
+
+def test_driver_root_env_override(monkeypatch):
+    monkeypatch.setenv("CRO_DRIVER_ROOT", "/custom/driver/root")
+    ex = MockNodeExec()
+    kfd_fixture(ex, 1)
+    ops = AmdNodeOps(ex, client=_container_driver_client("/ignored"))
+    assert ops.driver_root(NODE) == "/custom/driver/root"
+    assert ops._module_argv(NODE, ["modprobe", "-r", "amdgpu"]) == [
+        "chroot", "/custom/driver/root", "modprobe", "-r", "amdgpu"]
+
+
+def test_exec_probe_container_arm(monkeypatch):
+    """The gfx950 exec-probe runs through the driver-container chroot in
+    container mode (nvidia-smi-through-chroot analog)."""
+    import json
+
+    from cro_amd.nodeops.probe import make_exec_probe_fn
+
+    monkeypatch.delenv("CRO_DRIVER_ROOT", raising=False)
+    ex = MockNodeExec()
+    ids = kfd_fixture(ex, 1)
+    ops = AmdNodeOps(ex, client=_container_driver_client())
+    probe_result = {"ok": True, "mfma_f32_exact": True}
+    ex.set_command(
+        ("chroot", "/run/amdgpu-driver", "croagent", "probe", "--bdf", "0000:03:00.0"),
+        (0, json.dumps(probe_result), ""),
+    )
+    probe = make_exec_probe_fn(
+        ex, NODE, argv_prefix_fn=lambda: ops.probe_argv_prefix(NODE))
+    gpu = ops.find_gpu(NODE, ids[0])
+    assert probe(gpu)["ok"] is True
+    assert ("run", NODE,
+            ("chroot", "/run/amdgpu-driver", "croagent", "probe", "--bdf", "0000:03:00.0")
+            ) in ex.calls
