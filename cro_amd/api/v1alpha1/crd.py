@@ -167,6 +167,9 @@ def composable_resource_crd() -> dict:
                                         "error": {"type": "string"},
                                         "device_id": {"type": "string"},
                                         "cdi_device_id": {"type": "string"},
+                                        # extension: resumable async-fabric
+                                        # wait (types.py ComposableResourceStatus)
+                                        "fabric_wait_started": {"type": "string"},
                                     },
                                 },
                             },

@@ -138,12 +138,20 @@ class ComposableResourceSpec(_Model):
 
 
 class ComposableResourceStatus(_Model):
-    """Parity: ComposableResourceStatus, composableresource_types.go:36-41."""
+    """Parity: ComposableResourceStatus, composableresource_types.go:36-41.
+
+    ``fabric_wait_started`` is an extension beyond the reference: the
+    RFC3339 start of an in-progress async fabric wait, persisted so a
+    restarted operator resumes the exponential poll at the max interval
+    instead of hammering the fabric from the base interval again (the
+    reference keeps no such state and re-polls at its fixed 30 s quantum).
+    """
 
     state: str = ""
     error: str = ""
     device_id: str = ""
     cdi_device_id: str = ""
+    fabric_wait_started: str = ""
 
 
 class ComposableResource(K8sObject):

@@ -270,3 +270,52 @@ def test_gc_on_node_deletion(mock_world):
     assert "not found" in got.status.error
     mock_world.resource_rec.reconcile("gpu-1")  # Deleting → finalizer off → gone
     assert mock_world.client.try_get(ComposableResource, "gpu-1") is None
+
+
+def test_fabric_wait_resumes_at_max_after_restart(mock_world):
+    """the exponential async-fabric wait must survive an
+    operator restart — status.fabric_wait_started is persisted on the first
+    Waiting, and a fresh reconciler (empty in-memory counters) resumes at
+    fabric_wait_max instead of re-ramping from the base interval."""
+    from cro_amd.controllers.composableresource import (
+        ComposableResourceReconciler,
+        ReconcileConfig,
+    )
+    from cro_amd.fabric import WaitingDeviceAttaching
+
+    seed(mock_world)
+    # fabric stuck composing: every add_resource says "still attaching"
+    def waiting_add(resource):
+        raise WaitingDeviceAttaching("composing")
+
+    mock_world.fabric.add_resource = waiting_add
+    rec = mock_world.resource_rec
+    rec.reconcile("gpu-1")  # None → Attaching
+    r1 = rec.reconcile("gpu-1")  # first Waiting
+    assert r1.requeue_after == rec.config.fabric_wait_base
+    got = mock_world.client.get(ComposableResource, "gpu-1")
+    assert got.status.fabric_wait_started  # persisted in status
+
+    r2 = rec.reconcile("gpu-1")
+    assert r2.requeue_after == rec.config.fabric_wait_base * 2  # ramping
+
+    # "restart": fresh reconciler over the same store, no in-memory state
+    rec2 = ComposableResourceReconciler(
+        mock_world.client, mock_world.adapter, mock_world.ops, ReconcileConfig()
+    )
+    r3 = rec2.reconcile("gpu-1")
+    assert r3.requeue_after == rec2.config.fabric_wait_max  # no re-ramp
+
+    # fabric completes → wait state cleared
+    from tests.conftest import mock_world as _unused  # noqa: F401
+
+    def ok_add(resource):
+        mock_world.ops.fabric_composed(resource.spec.target_node, "GPU-done")
+        return "GPU-done", "amd.com/gpu=GPU-done"
+
+    mock_world.fabric.add_resource = ok_add
+    for _ in range(5):
+        rec2.reconcile("gpu-1")
+    got = mock_world.client.get(ComposableResource, "gpu-1")
+    assert got.status.state == "Online"
+    assert got.status.fabric_wait_started == ""

@@ -43,7 +43,10 @@ def test_resource_crd_schema_parity():
     assert set(spec_schema["required"]) == {"type", "model", "target_node"}
     assert spec_schema["properties"]["type"]["enum"] == ["gpu", "cxlmemory"]
     status = crd["spec"]["versions"][0]["schema"]["openAPIV3Schema"]["properties"]["status"]
-    assert set(status["properties"]) == {"state", "error", "device_id", "cdi_device_id"}
+    # reference fields + the documented fabric_wait_started extension
+    # (resumable async-fabric wait, types.py ComposableResourceStatus)
+    assert set(status["properties"]) == {
+        "state", "error", "device_id", "cdi_device_id", "fabric_wait_started"}
 
 
 def test_committed_manifests_match_generator():
