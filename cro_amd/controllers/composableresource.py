@@ -165,14 +165,20 @@ class ComposableResourceReconciler(Reconciler):
             self._fabric_polls[uid] = 17
             return self.config.fabric_wait_max
         self._fabric_polls[uid] = n + 1
-        if n == 0:
+        wait = min(self.config.fabric_wait_base * (2 ** min(n, 16)),
+                   self.config.fabric_wait_max)
+        # Persist the marker only once the ramp hits its cap: short composes
+        # finish before this and keep a write-free poll schedule (a status
+        # write would requeue us immediately and skew the ramp); a compose
+        # still pending at the cap is the long-running case where restart
+        # resumption matters.
+        if wait >= self.config.fabric_wait_max and not resource.status.fabric_wait_started:
             resource.status.fabric_wait_started = _now_rfc3339()
             try:
                 self.client.update_status(resource)
             except ConflictError:
                 pass  # a racing write wins; next poll persists again
-        return min(self.config.fabric_wait_base * (2 ** min(n, 16)),
-                   self.config.fabric_wait_max)
+        return wait
 
     def _persist_device_identity(
         self, name: str, device_id: str, cdi_device_id: str
