@@ -3,7 +3,7 @@
 PYTHON ?= python3
 HIPCC ?= /opt/rocm/bin/hipcc
 
-.PHONY: all build test test-gpu bench manifests lint docker-build clean
+.PHONY: all build test test-gpu bench manifests lint docker-build bundle clean
 
 all: build
 
@@ -31,6 +31,11 @@ lint:
 
 docker-build:
 	docker build -t cro-amd-operator:latest .
+
+# OLM bundle (operator-sdk `make bundle` analog, offline): assembles the
+# CSV base + CRDs + samples into bundle/ in registry+v1 layout.
+bundle: manifests
+	$(PYTHON) tools/make_bundle.py bundle
 
 clean:
 	rm -f cro_amd/hip/*.so
