@@ -95,6 +95,10 @@ def main(argv=None) -> int:
                    "analog). Ignored with --api-server.")
     p.add_argument("--destructive", action="store_true",
                    help="perform real PCI remove/rescan and module unload")
+    p.add_argument("--simulate-node-path", action="store_true",
+                   help="use the in-memory node-path double instead of real "
+                   "KFD/sysfs (CI and control-plane demos on GPU-less "
+                   "machines; BASELINE config #1 shape)")
     p.add_argument("--max-concurrent-reconciles", type=int, default=8)
     p.add_argument("--syncer-period", type=float, default=60.0)
     p.add_argument("--zap-log-level", default="info")
@@ -218,24 +222,42 @@ def main(argv=None) -> int:
     except Exception:
         log.warning("gfx950 probe library unavailable; health probe disabled")
     execer = LocalNodeExec()
-    gpu_ops = AmdNodeOps(
-        execer,
-        client=mgr.client,
-        cdi_dir=args.cdi_dir,
-        destructive=args.destructive,
-        probe_fn=probe_fn,
-    )
-    from ..nodeops.composite import CompositeNodeOps
-    from ..nodeops.cxl import CxlNodeOps
+    if args.simulate_node_path:
+        from ..nodeops.amdgpu import MockNodeOps
 
-    node_ops = CompositeNodeOps(
-        {
-            "gpu": gpu_ops,
-            "cxlmemory": CxlNodeOps(
-                execer, cdi_dir=args.cdi_dir, destructive=args.destructive
-            ),
-        }
-    )
+        gpu_ops = MockNodeOps(client=mgr.client)
+        node_ops = gpu_ops
+        # bridge: fabric composition makes the device node-visible after
+        # the rescan analog (the bench-harness binding)
+        provider = adapter.provider
+        orig_add = getattr(provider, "add_resource", None)
+        if orig_add is not None:
+            def _sim_add(resource, _orig=orig_add, _ops=gpu_ops):
+                did, cdi = _orig(resource)
+                _ops.fabric_composed(resource.spec.target_node, did)
+                return did, cdi
+
+            provider.add_resource = _sim_add
+        log.info("node path SIMULATED (--simulate-node-path)")
+    else:
+        gpu_ops = AmdNodeOps(
+            execer,
+            client=mgr.client,
+            cdi_dir=args.cdi_dir,
+            destructive=args.destructive,
+            probe_fn=probe_fn,
+        )
+        from ..nodeops.composite import CompositeNodeOps
+        from ..nodeops.cxl import CxlNodeOps
+
+        node_ops = CompositeNodeOps(
+            {
+                "gpu": gpu_ops,
+                "cxlmemory": CxlNodeOps(
+                    execer, cdi_dir=args.cdi_dir, destructive=args.destructive
+                ),
+            }
+        )
     mgr.resource_reconciler.node_ops = node_ops
     if hasattr(mgr, "syncer"):
         mgr.syncer.node_ops = node_ops
