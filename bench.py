@@ -163,6 +163,23 @@ def main() -> int:
         attach_ms = sorted(s["attach_ms"] for s in samples)
         detach_ms = sorted(s["detach_ms"] for s in samples)
 
+        # attach-phase breakdown (CRO_BENCH_PHASES=<path>) — side file so
+        # stdout stays the single contract JSON line
+        phases_path = os.environ.get("CRO_BENCH_PHASES", "")
+        if phases_path:
+            phases = {}
+            for metric in stack.mgr.metrics.attach_phase_seconds.collect():
+                for s in metric.samples:
+                    if s.name.endswith("_sum"):
+                        phases.setdefault(s.labels["phase"], {})["sum_s"] = s.value
+                    elif s.name.endswith("_count"):
+                        phases.setdefault(s.labels["phase"], {})["count"] = s.value
+            for v in phases.values():
+                if v.get("count"):
+                    v["avg_ms"] = round(v["sum_s"] * 1e3 / v["count"], 3)
+            with open(phases_path, "w") as f:
+                json.dump(phases, f, indent=2)
+
         extras = {}
         if not args.skip_extras:
             # pure per-device overhead path (round-1 headline shape)
