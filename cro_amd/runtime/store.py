@@ -331,11 +331,16 @@ class InMemoryStore:
             return obj.clone()
 
     def get(self, kind: str, name: str) -> K8sObject:
+        # stored objects are immutable-after-insert (every mutator swaps in
+        # a fresh object), so the snapshot ref is taken under the lock and
+        # the copy happens OUTSIDE it — clone cost never extends the
+        # store's critical section (the contended-bench queueing term)
         with self._lock:
             try:
-                return self._objects[kind][name].clone()
+                stored = self._objects[kind][name]
             except KeyError:
                 raise NotFoundError(f"{kind}/{name} not found") from None
+        return stored.clone()
 
     def list(
         self,
@@ -349,15 +354,15 @@ class InMemoryStore:
         O(n²) term). Callers must not mutate them."""
         with self._lock:
             items = list(self._objects.get(kind, {}).values())
-            if label_selector:
-                items = [
-                    o
-                    for o in items
-                    if all(o.metadata.labels.get(k) == v for k, v in label_selector.items())
-                ]
-            if not copy:
-                return items
-            return [o.clone() for o in items]
+        if label_selector:
+            items = [
+                o
+                for o in items
+                if all(o.metadata.labels.get(k) == v for k, v in label_selector.items())
+            ]
+        if not copy:
+            return items
+        return [o.clone() for o in items]  # outside the lock (see get)
 
     def update(self, obj: K8sObject) -> K8sObject:
         """Update metadata+spec; status is preserved from the stored object."""
@@ -425,10 +430,13 @@ class InMemoryStore:
             stored = self._require(kind, name)
             if stored.metadata.finalizers:
                 if stored.metadata.deletionTimestamp is None:
-                    old = stored.clone()
-                    stored.metadata.deletionTimestamp = _now_rfc3339()
-                    stored.metadata.resourceVersion = str(next(self._rv))
-                    self._notify(WatchEvent(MODIFIED, stored, old))
+                    # swap, never mutate in place: get/list hand out refs
+                    # and clone outside the lock (immutable-after-insert)
+                    new = stored.clone()
+                    new.metadata.deletionTimestamp = _now_rfc3339()
+                    new.metadata.resourceVersion = str(next(self._rv))
+                    self._objects[kind][name] = new
+                    self._notify(WatchEvent(MODIFIED, new, stored))
                 return
             self._finalize_delete(stored)
 
@@ -455,7 +463,10 @@ class InMemoryStore:
         del self._objects[stored.kind][stored.metadata.name]
         # a delete is a write: it gets its own resourceVersion, and the
         # DELETED event's object carries it (etcd-revision semantics — the
-        # resume-token space stays in lockstep with object RVs)
-        stored.metadata.resourceVersion = str(next(self._rv))
-        self._notify(WatchEvent(DELETED, stored))
-        return stored.clone()
+        # resume-token space stays in lockstep with object RVs). Copy
+        # before bumping: earlier get/list snapshots may still alias
+        # ``stored`` (immutable-after-insert contract).
+        final = stored.clone()
+        final.metadata.resourceVersion = str(next(self._rv))
+        self._notify(WatchEvent(DELETED, final))
+        return final
