@@ -279,3 +279,31 @@ def test_churn_races_on_real_node_path():
         assert stack.ops.cdi.devices("gpuchurn") == []
     finally:
         stack.mgr.stop()
+
+
+@pytest.mark.gpu
+@pytest.mark.timeout(300)
+def test_full_lifecycle_device_plugin_mode():
+    """The DEVICE_PLUGIN half of the hardware matrix: same real node path
+    (KFD, CDI, probe) with node-wide load checks and plugin-daemonset
+    refresh semantics instead of DRA ResourceSlices/taints."""
+    _require_gpu()
+    from cro_amd.api.v1alpha1.types import DeviceTaintRule, ResourceSlice
+    from cro_amd.bench_harness import attach_detach_cycle, build_local_stack
+
+    cdi_dir = os.path.join(os.environ.get("TMPDIR", "/tmp"), "cro-cdi-gputest-dp")
+    stack = build_local_stack(
+        node_name="gputest-dp", use_gpu=True, gpu_index=0,
+        cdi_dir=cdi_dir, mode="DEVICE_PLUGIN",
+    )
+    stack.mgr.start()
+    try:
+        timing = attach_detach_cycle(
+            stack, "gpu-e2e-dp", size=1, timeout=180, force_detach=True,
+        )
+        assert timing["attach_ms"] < 30000, timing
+        assert stack.ops.cdi.devices("gputest-dp") == []
+        # DEVICE_PLUGIN mode must not have used the DRA reflection kinds
+        assert stack.mgr.client.list(DeviceTaintRule) == []
+    finally:
+        stack.mgr.stop()
