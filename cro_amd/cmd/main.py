@@ -190,7 +190,38 @@ def main(argv=None) -> int:
     if args.api_server:
         from ..runtime.remote import RemoteClient
 
-        remote = RemoteClient(args.api_server)
+        remote = RemoteClient(args.api_server, token=api_token)
+        # fail fast on auth/connectivity: a token mismatch (each process
+        # auto-generates its own when the Secret is not set) would
+        # otherwise surface only as silent 401 watch-reconnect loops
+        from ..api.v1alpha1.types import Node
+        from ..runtime.errors import ApiError
+
+        deadline = 30.0
+        import time as _time
+
+        t0 = _time.monotonic()
+        while True:
+            try:
+                remote.list(Node)
+                break
+            except ApiError as exc:
+                if "401" in str(exc):
+                    log.error(
+                        "API server %s rejected our bearer token (401). Set "
+                        "CRO_API_TOKEN identically on both processes (the "
+                        "cro-amd-tokens Secret); auto-generated tokens are "
+                        "per-process.", args.api_server)
+                    return 1
+                if _time.monotonic() - t0 > deadline:
+                    log.error("API server %s unreachable: %s", args.api_server, exc)
+                    return 1
+                _time.sleep(0.5)
+            except Exception as exc:
+                if _time.monotonic() - t0 > deadline:
+                    log.error("API server %s unreachable: %s", args.api_server, exc)
+                    return 1
+                _time.sleep(0.5)
     store = None
     if remote is None and args.data_dir:
         from ..runtime.store import InMemoryStore

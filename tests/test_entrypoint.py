@@ -508,3 +508,50 @@ def test_boot_with_fti_fm_backend(tmp_path):
         except subprocess.TimeoutExpired:
             proc.kill()
     assert proc.returncode == 0
+
+
+@pytest.mark.timeout(120)
+def test_split_mode_token_mismatch_fails_fast(tmp_path):
+    """--api-server with a wrong bearer token must exit with a clear error
+    immediately, not loop on silent 401 watch reconnects."""
+    port = free_port()
+    env = dict(os.environ)
+    env.update({"DEVICE_RESOURCE_TYPE": "DRA", "CDI_PROVIDER_TYPE": "MOCK",
+                "CRO_API_TOKEN": "right-token"})
+    server = subprocess.Popen(
+        [sys.executable, "-m", "cro_amd.cmd.main",
+         "--api-bind-address", f":{port}",
+         "--metrics-bind-address", f":{free_port()}",
+         "--health-probe-bind-address", f":{free_port()}",
+         "--serve-only"],
+        cwd=REPO, env=env,
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+    )
+    try:
+        deadline = time.monotonic() + 30
+        while time.monotonic() < deadline:
+            try:
+                if httpx.get(f"http://127.0.0.1:{port}/healthz", timeout=1).status_code == 200:
+                    break
+            except Exception:
+                time.sleep(0.2)
+        env2 = dict(env)
+        env2["CRO_API_TOKEN"] = "wrong-token"
+        op = subprocess.run(
+            [sys.executable, "-m", "cro_amd.cmd.main",
+             "--api-server", f"http://127.0.0.1:{port}",
+             "--api-bind-address", f":{free_port()}",
+             "--metrics-bind-address", f":{free_port()}",
+             "--health-probe-bind-address", f":{free_port()}",
+             "--simulate-node-path"],
+            cwd=REPO, env=env2,
+            capture_output=True, text=True, timeout=60,
+        )
+        assert op.returncode == 1
+        assert "rejected our bearer token" in op.stdout + op.stderr
+    finally:
+        server.send_signal(signal.SIGTERM)
+        try:
+            server.wait(timeout=15)
+        except subprocess.TimeoutExpired:
+            server.kill()
