@@ -360,7 +360,9 @@ def main(argv=None) -> int:
     import uvicorn
 
     app = build_app(mgr.client, token=api_token)
-    if args.node:  # node-agent surface for off-node controllers
+    if args.node and not args.simulate_node_path:
+        # node-agent surface for off-node controllers (the simulated node
+        # path has no NodeExec to expose)
         build_agent_app(
             node_ops.execer, node_name=args.node, app=app, token=agent_token
         )
