@@ -134,6 +134,14 @@ def main(argv=None) -> int:
             level=getattr(logging, args.zap_log_level.upper(), logging.INFO),
             format="%(asctime)s %(levelname)s %(name)s %(message)s",
         )
+    # httpx/httpcore log every request at INFO — in remote mode that is
+    # 30+ lines per reconcile cycle of pure noise (and enough volume to
+    # block the process if a supervisor attaches a pipe it never drains).
+    # Production norm: transport logs only at WARNING unless debugging.
+    if args.zap_log_level.lower() != "debug":
+        for noisy in ("httpx", "httpcore"):
+            logging.getLogger(noisy).setLevel(logging.WARNING)
+
     log = logging.getLogger("cro_amd.main")
 
     # -- fail-closed bearer tokens ------------------------------------------

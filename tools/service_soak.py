@@ -60,10 +60,16 @@ def main() -> int:
             "--cdi-dir", os.path.join(os.environ.get("TMPDIR", "/tmp"), "cro-soak-cdi")]
     if not os.path.exists("/dev/kfd"):
         argv.append("--simulate-node-path")  # GPU-less dry runs
+    import tempfile
+
+    # file-backed stdout: a PIPE nobody drains blocks the child once the
+    # 64 KB buffer fills under steady logging (found by split_soak.py)
+    logf = tempfile.NamedTemporaryFile(
+        mode="w+", prefix="service-soak-", suffix=".log", delete=False)
     proc = subprocess.Popen(
         argv,
         cwd=REPO, env=env,
-        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+        stdout=logf, stderr=subprocess.STDOUT, text=True,
     )
     http = httpx.Client(
         base_url=f"http://127.0.0.1:{port}", timeout=10,
@@ -126,10 +132,13 @@ def main() -> int:
     finally:
         proc.send_signal(signal.SIGTERM)
         try:
-            out, _ = proc.communicate(timeout=20)
+            proc.wait(timeout=20)
         except subprocess.TimeoutExpired:
             proc.kill()
-            out, _ = proc.communicate()
+            proc.wait()
+        logf.flush()
+        logf.seek(0)
+        out = logf.read()
 
     err_lines = [l for l in out.splitlines()
                  if "ERROR" in l or "Traceback" in l]
