@@ -198,7 +198,13 @@ def main(argv=None) -> int:
     if args.api_server:
         from ..runtime.remote import RemoteClient
 
-        remote = RemoteClient(args.api_server, token=api_token)
+        # SharedInformer read cache (client-go shape): reconciles read from
+        # the watch-fed local cache instead of paying a GET RTT per read.
+        # CRO_CLIENT_CACHE=off falls back to direct reads.
+        remote = RemoteClient(
+            args.api_server, token=api_token,
+            cache=os.environ.get("CRO_CLIENT_CACHE", "on") != "off",
+        )
         # fail fast on auth/connectivity: a token mismatch (each process
         # auto-generates its own when the Secret is not set) would
         # otherwise surface only as silent 401 watch-reconnect loops

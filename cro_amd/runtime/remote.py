@@ -101,7 +101,15 @@ class RemoteClient:
         token: Optional[str] = None,
         ca: Optional[str] = None,
         verify: Union[bool, str, None] = None,
+        cache: bool = False,
     ):
+        """``cache=True`` enables the client-go SharedInformer read path:
+        one list+watch informer per kind maintains a local object cache;
+        ``get``/``list`` are served from it once synced (a reconcile then
+        costs zero read RTTs, exactly as controller-runtime reads from its
+        cache), writes go to the server and update the cache read-your-
+        writes-style. Reads may be marginally stale — level-triggered
+        reconciles + optimistic concurrency absorb that, as in k8s."""
         import os
 
         self.base_url = base_url.rstrip("/")
@@ -118,6 +126,20 @@ class RemoteClient:
         )
         self._stop = threading.Event()
         self._watch_threads: List[threading.Thread] = []
+        self._cache_enabled = cache
+        self._informers: Dict[str, "_Informer"] = {}
+        self._informer_lock = threading.Lock()
+
+    # -- informer cache ----------------------------------------------------
+
+    def _informer(self, kind: str) -> "_Informer":
+        with self._informer_lock:
+            inf = self._informers.get(kind)
+            if inf is None:
+                inf = _Informer(self, kind)
+                self._informers[kind] = inf
+                inf.start()
+            return inf
 
     # -- kind plumbing -----------------------------------------------------
 
@@ -135,10 +157,28 @@ class RemoteClient:
         cls, plural = self._resolve(type(obj))
         resp = self._http.post(f"{BASE}/{plural}", json=obj.model_dump(by_alias=True))
         _raise_for(resp)
-        return cls.model_validate(resp.json())
+        created = cls.model_validate(resp.json())
+        self._offer_cache(created)
+        return created
+
+    def _offer_cache(self, obj) -> None:
+        """Read-your-writes: fold a write response into the informer cache
+        (kept only if newer than the cached rv)."""
+        if not self._cache_enabled:
+            return
+        inf = self._informers.get(obj.kind)
+        if inf is not None:
+            inf.offer(obj)
 
     def get(self, cls_or_kind, name: str):
         cls, plural = self._resolve(cls_or_kind)
+        if self._cache_enabled:
+            inf = self._informer(cls.KIND)
+            if inf.synced.is_set():
+                cached = inf.get_cached(name)
+                if cached is None:
+                    raise NotFoundError(f"{cls.KIND}/{name} not found")
+                return cached
         resp = self._http.get(f"{BASE}/{plural}/{name}")
         _raise_for(resp)
         return cls.model_validate(resp.json())
@@ -151,6 +191,10 @@ class RemoteClient:
 
     def list(self, cls_or_kind, labels: Optional[Dict[str, str]] = None, copy: bool = True):
         cls, plural = self._resolve(cls_or_kind)
+        if self._cache_enabled:
+            inf = self._informer(cls.KIND)
+            if inf.synced.is_set():
+                return inf.list_cached(labels)
         params = {}
         if labels:
             params["labelSelector"] = ",".join(f"{k}={v}" for k, v in labels.items())
@@ -164,7 +208,9 @@ class RemoteClient:
             f"{BASE}/{plural}/{obj.metadata.name}", json=obj.model_dump(by_alias=True)
         )
         _raise_for(resp)
-        return cls.model_validate(resp.json())
+        updated = cls.model_validate(resp.json())
+        self._offer_cache(updated)
+        return updated
 
     def update_status(self, obj: T) -> T:
         cls, plural = self._resolve(type(obj))
@@ -173,7 +219,9 @@ class RemoteClient:
             json=obj.model_dump(by_alias=True),
         )
         _raise_for(resp)
-        return cls.model_validate(resp.json())
+        updated = cls.model_validate(resp.json())
+        self._offer_cache(updated)
+        return updated
 
     def delete(self, obj_or_cls, name: Optional[str] = None) -> None:
         if name is None:
@@ -189,21 +237,30 @@ class RemoteClient:
     def watch(self, kinds: Optional[List[str]] = None) -> "queue.Queue[WatchEvent]":
         q: "queue.Queue[WatchEvent]" = queue.Queue()
         for kind in kinds or list(_PLURALS):
-            t = threading.Thread(
-                target=self._watch_kind, args=(kind, q),
-                name=f"remote-watch-{kind}", daemon=True,
-            )
-            t.start()
-            self._watch_threads.append(t)
+            if self._cache_enabled:
+                # SharedInformer shape: one stream per kind, subscribers
+                # get a cache replay + the shared live feed
+                self._informer(kind).subscribe(q)
+            else:
+                t = threading.Thread(
+                    target=self._watch_kind, args=(kind, q),
+                    name=f"remote-watch-{kind}", daemon=True,
+                )
+                t.start()
+                self._watch_threads.append(t)
         return q
 
-    def _watch_kind(self, kind: str, q: "queue.Queue[WatchEvent]") -> None:
+    def _watch_kind(self, kind: str, q, on_event=None, on_synced=None) -> None:
         """The informer protocol, exactly as client-go runs it against a
         kube-apiserver: LIST (take ListMeta.resourceVersion) → synthesize
         ADDED for the current objects (cache replay) → WATCH from that rv
         with bookmarks → on 410 Expired (ERROR event) or an aged-out
         reconnect, re-list.  Resume tokens are object resourceVersions —
-        there is no out-of-band framing field."""
+        there is no out-of-band framing field.
+
+        ``on_event(WatchEvent)`` (default: ``q.put``) receives every event;
+        ``on_synced()`` fires after each initial list completes."""
+        deliver = on_event if on_event is not None else q.put
         cls, plural = self._resolve(kind)
         last_rv = -1
         while not self._stop.is_set():
@@ -214,7 +271,9 @@ class RemoteClient:
                     body = resp.json()
                     last_rv = int(body.get("metadata", {}).get("resourceVersion", 0))
                     for item in body["items"]:
-                        q.put(WatchEvent("ADDED", cls.model_validate(item)))
+                        deliver(WatchEvent("ADDED", cls.model_validate(item)))
+                    if on_synced is not None:
+                        on_synced()
                 params = {
                     "watch": "true",
                     "resourceVersion": str(last_rv),
@@ -242,7 +301,7 @@ class RemoteClient:
                             last_rv = int(obj_rv)
                         if ev["type"] == "BOOKMARK":
                             continue  # resume-token refresh only
-                        q.put(WatchEvent(ev["type"], cls.model_validate(ev["object"])))
+                        deliver(WatchEvent(ev["type"], cls.model_validate(ev["object"])))
             except Exception as exc:
                 if self._stop.is_set():
                     return
@@ -252,3 +311,100 @@ class RemoteClient:
     def close(self) -> None:
         self._stop.set()
         self._http.close()
+
+
+class _Informer:
+    """client-go SharedInformer analog: ONE list+watch stream per kind
+    feeding a local object cache plus any number of subscriber queues.
+    Reads from the cache cost zero RTTs; subscribers joining later get a
+    replay of the current cache before live events."""
+
+    def __init__(self, client: "RemoteClient", kind: str):
+        self.client = client
+        self.kind = kind
+        self.synced = threading.Event()
+        self._lock = threading.Lock()
+        self._store: Dict[str, object] = {}
+        # deletion tombstones: name -> rv of the DELETED event. A write
+        # RESPONSE being folded in (offer) can race the watch thread's
+        # DELETED and resurrect a dead object without this (observed as a
+        # Cleaning loop forever deleting a phantom child).
+        self._tombstones: Dict[str, int] = {}
+        self._subs: List["queue.Queue[WatchEvent]"] = []
+        self._thread: Optional[threading.Thread] = None
+
+    def start(self) -> None:
+        self._thread = threading.Thread(
+            target=self.client._watch_kind,
+            args=(self.kind, None),
+            kwargs={"on_event": self._apply, "on_synced": self.synced.set},
+            name=f"informer-{self.kind}",
+            daemon=True,
+        )
+        self._thread.start()
+        self.client._watch_threads.append(self._thread)
+
+    # -- cache maintenance ---------------------------------------------------
+
+    @staticmethod
+    def _rv(obj) -> int:
+        try:
+            return int(obj.metadata.resourceVersion)
+        except (TypeError, ValueError):
+            return 0
+
+    def _apply(self, ev: WatchEvent) -> None:
+        name = ev.object.metadata.name
+        with self._lock:
+            if ev.type == "DELETED":
+                self._store.pop(name, None)
+                self._tombstones[name] = self._rv(ev.object)
+                if len(self._tombstones) > 8192:  # bound (names are uuids)
+                    for old in sorted(self._tombstones, key=self._tombstones.get)[:4096]:
+                        del self._tombstones[old]
+            else:
+                rv = self._rv(ev.object)
+                if rv > self._tombstones.get(name, -1):  # re-created object
+                    self._tombstones.pop(name, None)
+                    cur = self._store.get(name)
+                    if cur is None or rv >= self._rv(cur):
+                        self._store[name] = ev.object
+            subs = list(self._subs)
+        for q in subs:
+            q.put(ev)
+
+    def offer(self, obj) -> None:
+        """Fold a WRITE RESPONSE into the cache (read-your-writes): kept
+        only if at least as new as the cached entry and not superseded by
+        a DELETED tombstone (one rv space makes the comparison exact).
+        No fan-out — the watch stream delivers the canonical event."""
+        name = obj.metadata.name
+        with self._lock:
+            if self._rv(obj) <= self._tombstones.get(name, -1):
+                return  # deleted after this write; do not resurrect
+            cur = self._store.get(name)
+            if cur is None or self._rv(obj) >= self._rv(cur):
+                self._store[name] = obj.clone()
+
+    # -- consumers -----------------------------------------------------------
+
+    def subscribe(self, q: "queue.Queue[WatchEvent]") -> None:
+        with self._lock:
+            for obj in self._store.values():
+                q.put(WatchEvent("ADDED", obj))
+            self._subs.append(q)
+
+    def get_cached(self, name: str):
+        with self._lock:
+            obj = self._store.get(name)
+        return obj.clone() if obj is not None else None
+
+    def list_cached(self, labels: Optional[Dict[str, str]] = None):
+        with self._lock:
+            items = list(self._store.values())
+        if labels:
+            items = [
+                o for o in items
+                if all(o.metadata.labels.get(k) == v for k, v in labels.items())
+            ]
+        return [o.clone() for o in items]

@@ -552,3 +552,100 @@ def test_openshift_machine_chain_over_remote(api_server):
         assert resolve_machine_id_openshift(remote, "worker-0") == "uuid-42"
     finally:
         remote.close()
+
+
+def test_cached_client_informer_reads(api_server):
+    """cache=True: get/list served from the watch-fed informer cache
+    (client-go SharedInformer shape) — reads cost zero RTTs once synced,
+    write responses land read-your-writes, deletes converge via the
+    watch."""
+    url, server_mgr = api_server
+    server_mgr.client.create(make_request("c1", target_node="node0"))
+
+    rc = RemoteClient(url, cache=True)
+    try:
+        # first read starts the informer (HTTP fallback until synced)
+        items = rc.list(ComposabilityRequest)
+        assert [i.metadata.name for i in items] == ["c1"]
+        inf = rc._informers["ComposabilityRequest"]
+        assert inf.synced.wait(10)
+        # synced: reads now come from the cache
+        assert [i.metadata.name for i in rc.list(ComposabilityRequest)] == ["c1"]
+
+        # server-side create propagates into the cache via the watch
+        server_mgr.client.create(make_request("c2", model="m2", target_node="node0"))
+        deadline = time.monotonic() + 10
+        while time.monotonic() < deadline:
+            if rc.try_get(ComposabilityRequest, "c2") is not None:
+                break
+            time.sleep(0.02)
+        assert rc.get(ComposabilityRequest, "c2").spec.resource.model == "m2"
+
+        # read-your-writes: an update is visible immediately (no watch wait)
+        cur = rc.get(ComposabilityRequest, "c1")
+        cur.spec.resource.size = 3
+        updated = rc.update(cur)
+        assert rc.get(ComposabilityRequest, "c1").spec.resource.size == 3
+        assert int(rc.get(ComposabilityRequest, "c1").metadata.resourceVersion) >= int(
+            updated.metadata.resourceVersion)
+
+        # cached reads are COPIES: mutating one does not poison the cache
+        a = rc.get(ComposabilityRequest, "c1")
+        a.spec.resource.size = 99
+        assert rc.get(ComposabilityRequest, "c1").spec.resource.size == 3
+
+        # delete converges through the watch stream
+        server_mgr.client.delete(ComposabilityRequest, "c2")
+        deadline = time.monotonic() + 10
+        while time.monotonic() < deadline:
+            if rc.try_get(ComposabilityRequest, "c2") is None:
+                break
+            time.sleep(0.02)
+        assert rc.try_get(ComposabilityRequest, "c2") is None
+    finally:
+        rc.close()
+
+
+def test_cached_client_runs_controllers(api_server):
+    """A full operator over the CACHED client drives a request to Running
+    — the controller-runtime topology (controllers read from cache)."""
+    url, server_mgr = api_server
+    remote = RemoteClient(url, cache=True)
+    fabric = MockFabric(models={"mi355x": 8})
+    mgr = build_manager(
+        Adapter("DRA", fabric), None, client=remote, enable_webhook=False
+    )
+    ops = MockNodeOps(client=mgr.client)
+    mgr.resource_reconciler.node_ops = ops
+
+    orig_add = fabric.add_resource
+
+    def add_resource(resource):
+        did, cdi = orig_add(resource)
+        ops.fabric_composed(resource.spec.target_node, did)
+        return did, cdi
+
+    fabric.add_resource = add_resource
+    node = Node()
+    node.metadata.name = "cachenode"
+    node.status.capacity.milli_cpu = 64000
+    node.status.capacity.memory = 1 << 40
+    node.status.capacity.allowed_pod_number = 128
+    server_mgr.client.create(node)
+    mgr.start()
+    try:
+        server_mgr.client.create(make_request("cr1", size=2, target_node="cachenode"))
+        assert mgr.wait_for(
+            lambda: (r := server_mgr.client.try_get(ComposabilityRequest, "cr1")) is not None
+            and r.status.state == "Running",
+            timeout=30,
+        ), (lambda r: r.status.state if r else "gone")(
+            server_mgr.client.try_get(ComposabilityRequest, "cr1"))
+        server_mgr.client.delete(ComposabilityRequest, "cr1")
+        assert mgr.wait_for(
+            lambda: server_mgr.client.try_get(ComposabilityRequest, "cr1") is None,
+            timeout=30,
+        )
+    finally:
+        mgr.stop()
+        remote.close()
