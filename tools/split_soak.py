@@ -150,9 +150,11 @@ def main() -> int:
                 failed += 1
     finally:
         outs = []
-        for proc in procs:
+        # operator first, apiserver second: the operator's recorder/
+        # controllers drain against a live API instead of logging
+        # connection-refused noise
+        for proc in reversed(procs):
             proc.send_signal(signal.SIGTERM)
-        for proc in procs:
             try:
                 proc.wait(timeout=20)
             except subprocess.TimeoutExpired:
