@@ -53,6 +53,12 @@ _COLUMNS["resourceslices"] = (
     ("DRIVER", lambda o: o["spec"]["driver"]),
     ("DEVICES", lambda o: str(len(o["spec"]["devices"]))),
 )
+_COLUMNS["leases"] = (
+    ("NAME", lambda o: o["metadata"]["name"]),
+    ("HOLDER", lambda o: o["spec"].get("holderIdentity", "") or "<released>"),
+    ("RENEWED", lambda o: (o["spec"].get("renewTime") or "")[11:19]),
+    ("TRANSITIONS", lambda o: str(o["spec"].get("leaseTransitions", 0))),
+)
 _COLUMNS["events"] = (
     ("LAST-SEEN", lambda o: (o.get("last_seen", "") or "")[11:19]),
     ("TYPE", lambda o: o.get("type", "")),
