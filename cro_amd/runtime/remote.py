@@ -308,6 +308,13 @@ class RemoteClient:
                 log.debug("watch %s disconnected (%s); reconnecting", kind, exc)
                 self._stop.wait(0.2)
 
+    def stop_watch(self, q) -> None:
+        """Unsubscribe a queue produced by watch() (cache mode): stopped
+        controllers must not keep receiving fan-out into dead queues.
+        Legacy (uncached) watch threads exit with close() instead."""
+        for inf in self._informers.values():
+            inf.unsubscribe(q)
+
     def close(self) -> None:
         self._stop.set()
         self._http.close()
@@ -393,6 +400,11 @@ class _Informer:
             for obj in self._store.values():
                 q.put(WatchEvent("ADDED", obj))
             self._subs.append(q)
+
+    def unsubscribe(self, q) -> None:
+        with self._lock:
+            if q in self._subs:
+                self._subs.remove(q)
 
     def get_cached(self, name: str):
         with self._lock:
