@@ -649,3 +649,61 @@ def test_cached_client_runs_controllers(api_server):
     finally:
         mgr.stop()
         remote.close()
+
+
+def test_cached_client_converges_under_random_ops(api_server):
+    """Property: after ANY sequence of server-side create/update/delete
+    (including rapid name reuse — the tombstone race surface), the
+    informer cache converges to exactly the server's object set."""
+    import random
+
+    url, server_mgr = api_server
+    rc = RemoteClient(url, cache=True)
+    rng = random.Random(7)
+    try:
+        rc.list(ComposabilityRequest)  # start + sync the informer
+        assert rc._informers["ComposabilityRequest"].synced.wait(10)
+
+        names = [f"p{i}" for i in range(6)]
+        live = {}
+        for step in range(300):
+            name = rng.choice(names)
+            op = rng.random()
+            if name not in live:
+                server_mgr.client.create(
+                    make_request(name, model=f"m-{name}", target_node="node0"))
+                live[name] = 1
+            elif op < 0.5:
+                cur = server_mgr.client.get(ComposabilityRequest, name)
+                cur.spec.resource.size = rng.randint(1, 8)
+                server_mgr.client.update(cur)
+            else:
+                server_mgr.client.delete(ComposabilityRequest, name)
+                del live[name]
+            if rng.random() < 0.2:
+                # interleave CLIENT-SIDE writes (the offer/tombstone path)
+                cached = rc.try_get(ComposabilityRequest, name)
+                if cached is not None:
+                    cached.spec.resource.size = rng.randint(1, 8)
+                    try:
+                        rc.update(cached)
+                    except Exception:
+                        pass  # conflict/deleted under us — expected
+
+        # convergence: cache == server (object set and resourceVersions)
+        deadline = time.monotonic() + 10
+        while time.monotonic() < deadline:
+            server_objs = {
+                o.metadata.name: o.metadata.resourceVersion
+                for o in server_mgr.client.list(ComposabilityRequest)
+            }
+            cache_objs = {
+                o.metadata.name: o.metadata.resourceVersion
+                for o in rc.list(ComposabilityRequest)
+            }
+            if server_objs == cache_objs:
+                break
+            time.sleep(0.05)
+        assert server_objs == cache_objs, (server_objs, cache_objs)
+    finally:
+        rc.close()
