@@ -114,6 +114,14 @@ def main() -> int:
         reconcile_count,
     )
 
+    # Multi-rank runs: sibling ranks hold HIP/NCCL contexts on the devices
+    # (torchrun protocol), which the detach load check would correctly
+    # flag as foreign compute processes. force_detach (a first-class spec
+    # field) skips that check for the bench job; the load check itself is
+    # validated at N=1 and by tests/test_gpu.py.
+    if world > 1:
+        args.force_detach = True
+
     result = None
     if rank == 0:
         node_name = "bench-node"
@@ -338,6 +346,7 @@ def main() -> int:
                 "node_path": "real KFD/CDI/HIP-probe" if use_gpu else "mock",
                 "device_resource_type": args.mode,
                 "probe": bool(use_gpu and not args.no_probe),
+                "force_detach": bool(args.force_detach),
             },
             **extras,
         }
