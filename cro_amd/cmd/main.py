@@ -388,6 +388,9 @@ def main(argv=None) -> int:
             log_level="warning",
             ssl_certfile=args.tls_cert_file or None,
             ssl_keyfile=args.tls_key_file or None,
+            # bound SIGTERM: open watch STREAMS otherwise keep graceful
+            # shutdown waiting until the watch timeout (minutes)
+            timeout_graceful_shutdown=5,
         )
     )
 
@@ -428,6 +431,7 @@ def main(argv=None) -> int:
                     log_level="warning",
                     ssl_certfile=cert if have_tls else None,
                     ssl_keyfile=key if have_tls else None,
+                    timeout_graceful_shutdown=5,
                 )
             )
             webhook_thread = threading.Thread(

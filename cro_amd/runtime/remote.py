@@ -259,7 +259,9 @@ class RemoteClient:
         there is no out-of-band framing field.
 
         ``on_event(WatchEvent)`` (default: ``q.put``) receives every event;
-        ``on_synced()`` fires after each initial list completes."""
+        ``on_synced(names)`` fires after each initial list completes with
+        the set of listed names — a cache re-syncing on reconnect prunes
+        entries deleted during the disconnect (client-go re-list sync)."""
         deliver = on_event if on_event is not None else q.put
         cls, plural = self._resolve(kind)
         last_rv = -1
@@ -270,10 +272,13 @@ class RemoteClient:
                     _raise_for(resp)
                     body = resp.json()
                     last_rv = int(body.get("metadata", {}).get("resourceVersion", 0))
+                    listed = set()
                     for item in body["items"]:
-                        deliver(WatchEvent("ADDED", cls.model_validate(item)))
+                        obj = cls.model_validate(item)
+                        listed.add(obj.metadata.name)
+                        deliver(WatchEvent("ADDED", obj))
                     if on_synced is not None:
-                        on_synced()
+                        on_synced(listed)
                 params = {
                     "watch": "true",
                     "resourceVersion": str(last_rv),
@@ -344,12 +349,29 @@ class _Informer:
         self._thread = threading.Thread(
             target=self.client._watch_kind,
             args=(self.kind, None),
-            kwargs={"on_event": self._apply, "on_synced": self.synced.set},
+            kwargs={"on_event": self._apply, "on_synced": self._sync_complete},
             name=f"informer-{self.kind}",
             daemon=True,
         )
         self._thread.start()
         self.client._watch_threads.append(self._thread)
+
+    def _sync_complete(self, listed_names) -> None:
+        """End of an initial list (first sync OR a 410 re-list): any cached
+        object absent from the fresh list was deleted while disconnected —
+        prune it and deliver the synthetic DELETED client-go would."""
+        stale_events = []
+        with self._lock:
+            for name in list(self._store):
+                if name not in listed_names:
+                    obj = self._store.pop(name)
+                    self._tombstones[name] = self._rv(obj)
+                    stale_events.append(WatchEvent("DELETED", obj))
+            subs = list(self._subs)
+        for ev in stale_events:
+            for q in subs:
+                q.put(ev)
+        self.synced.set()
 
     # -- cache maintenance ---------------------------------------------------
 

@@ -65,7 +65,8 @@ class HealthServer:
             return {"status": "ok"}
 
         self.server = uvicorn.Server(
-            uvicorn.Config(app, host=host, port=port, log_level="warning")
+            uvicorn.Config(app, host=host, port=port, log_level="warning",
+                           timeout_graceful_shutdown=5)
         )
         self._thread: Optional[threading.Thread] = None
 
@@ -115,6 +116,7 @@ class MetricsServer:
                 log_level="warning",
                 ssl_certfile=certfile,
                 ssl_keyfile=keyfile,
+                timeout_graceful_shutdown=5,
             )
         )
         self._thread: Optional[threading.Thread] = None

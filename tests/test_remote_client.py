@@ -707,3 +707,84 @@ def test_cached_client_converges_under_random_ops(api_server):
         assert server_objs == cache_objs, (server_objs, cache_objs)
     finally:
         rc.close()
+
+
+def test_cached_client_relist_prunes_deleted():
+    """An object deleted while the informer is disconnected must be pruned
+    by the reconnect re-list (client-go sync semantics) — otherwise the
+    cache serves phantoms forever. Forced via apiserver replacement on the
+    same port with a store missing one object (state loss → foreign resume
+    rv → 410 Expired → re-list → prune + synthetic DELETED).  A short
+    CRO_WATCH_TIMEOUT lets the first server's stream end so it can shut
+    down (open streams otherwise pin uvicorn's graceful shutdown)."""
+    import os as _os
+
+    import uvicorn
+
+    _os.environ["CRO_WATCH_TIMEOUT"] = "2"
+    port = free_port()
+
+    def start_server(mgr):
+        server = uvicorn.Server(uvicorn.Config(
+            build_app(mgr.client), host="127.0.0.1", port=port,
+            log_level="error"))
+        thread = threading.Thread(target=server.run, daemon=True)
+        thread.start()
+        import httpx
+
+        deadline = time.monotonic() + 15
+        while time.monotonic() < deadline:
+            try:
+                if httpx.get(f"http://127.0.0.1:{port}/healthz", timeout=1).status_code == 200:
+                    return server, thread
+            except Exception:
+                time.sleep(0.05)
+        raise AssertionError("server did not come up")
+
+    mgr1 = build_manager(Adapter("DRA", MockFabric()), None)
+    mgr1.client.create(make_request("keep", target_node="node0"))
+    mgr1.client.create(make_request("vanish", model="m2", target_node="node0"))
+    server1, thread1 = start_server(mgr1)
+
+    rc = RemoteClient(f"http://127.0.0.1:{port}", cache=True)
+    try:
+        rc.list(ComposabilityRequest)
+        inf = rc._informers["ComposabilityRequest"]
+        assert inf.synced.wait(10)
+        assert wait_for(
+            lambda: {o.metadata.name for o in rc.list(ComposabilityRequest)}
+            == {"keep", "vanish"}, timeout=10)
+        q = rc.watch(["ComposabilityRequest"])
+        while not q.empty():
+            q.get_nowait()  # drain the subscribe replay
+
+        server1.should_exit = True
+        thread1.join(timeout=5)
+
+        mgr2 = build_manager(Adapter("DRA", MockFabric()), None)
+        mgr2.client.create(make_request("keep", target_node="node0"))
+        server2, thread2 = start_server(mgr2)
+        try:
+            # the informer reconnects, re-lists, prunes "vanish"
+            assert wait_for(
+                lambda: {o.metadata.name for o in rc.list(ComposabilityRequest)}
+                == {"keep"}, timeout=20), [
+                    o.metadata.name for o in rc.list(ComposabilityRequest)]
+            assert rc.try_get(ComposabilityRequest, "vanish") is None
+            # the subscriber saw the synthetic DELETED
+            deadline = time.monotonic() + 10
+            deleted = []
+            while time.monotonic() < deadline and not deleted:
+                try:
+                    ev = q.get(timeout=1)
+                except Exception:
+                    continue
+                if ev.type == "DELETED" and ev.object.metadata.name == "vanish":
+                    deleted.append(ev)
+            assert deleted, "no synthetic DELETED for the pruned object"
+        finally:
+            server2.should_exit = True
+            thread2.join(timeout=5)
+    finally:
+        _os.environ.pop("CRO_WATCH_TIMEOUT", None)
+        rc.close()
