@@ -178,6 +178,7 @@ def build_app(client: Client, token: str = None) -> FastAPI:
     async def list_objects(
         plural: str,
         labelSelector: str = "",
+        fieldSelector: str = "",
         watch: bool = False,
         resourceVersion: str = "",
         allowWatchBookmarks: bool = False,
@@ -194,6 +195,16 @@ def build_app(client: Client, token: str = None) -> FastAPI:
         if labelSelector:
             labels = dict(part.split("=", 1) for part in labelSelector.split(","))
         items = client.list(cls, labels)
+        if fieldSelector:
+            # the metadata.name form kubectl/client-go use (equality only,
+            # matching the k8s supported fields for custom resources)
+            for part in fieldSelector.split(","):
+                key, _, value = part.partition("=")
+                if key == "metadata.name":
+                    items = [o for o in items if o.metadata.name == value]
+                else:
+                    return _error_response(ValueError(
+                        f"unsupported fieldSelector {key!r} (metadata.name only)"))
         # ListMeta.resourceVersion is the watch-resume token for the
         # list-then-watch informer protocol (apiserver parity)
         return {

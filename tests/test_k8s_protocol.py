@@ -317,3 +317,22 @@ def test_remote_client_list_then_watch(stack):
         rc.close()
         server.should_exit = True
         t.join(timeout=5)
+
+
+def test_field_selector_metadata_name(stack):
+    """kubectl's fieldSelector=metadata.name=<x> form (the supported field
+    for custom resources); unsupported fields get a 422 Status."""
+    http, mgr = stack
+    mgr.client.create(make_request("fs1", target_node="n0"))
+    mgr.client.create(make_request("fs2", model="m2", target_node="n0"))
+    body = http.get(
+        f"{GROUP}/composabilityrequests",
+        params={"fieldSelector": "metadata.name=fs2"},
+    ).json()
+    assert [i["metadata"]["name"] for i in body["items"]] == ["fs2"]
+    resp = http.get(
+        f"{GROUP}/composabilityrequests",
+        params={"fieldSelector": "spec.nodeName=x"},
+    )
+    assert resp.status_code == 422
+    assert resp.json()["kind"] == "Status"
