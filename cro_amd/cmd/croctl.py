@@ -118,6 +118,12 @@ def main(argv=None, client: httpx.Client = None) -> int:
     rm.add_argument("plural")
     rm.add_argument("name")
 
+    pa = sub.add_parser("patch")
+    pa.add_argument("plural")
+    pa.add_argument("name")
+    pa.add_argument("-p", "--patch", required=True,
+                    metavar="JSON", help="RFC 7386 merge patch body")
+
     sc = sub.add_parser("scale")
     sc.add_argument("plural")
     sc.add_argument("name")
@@ -286,6 +292,17 @@ def main(argv=None, client: httpx.Client = None) -> int:
             if resp.status_code not in (200, 201):
                 return fail(resp)
             print(f"{plural}/{name} {verb}")
+        return 0
+
+    if args.command == "patch":
+        resp = http.patch(
+            f"{BASE}/{args.plural}/{args.name}",
+            content=args.patch,
+            headers={"Content-Type": "application/merge-patch+json"},
+        )
+        if resp.status_code != 200:
+            return fail(resp)
+        print(f"{args.plural}/{args.name} patched")
         return 0
 
     if args.command == "delete":

@@ -69,6 +69,22 @@ def _dump(obj) -> dict:
     return obj.model_dump(by_alias=True)
 
 
+def _merge_patch(target, patch):
+    """RFC 7386 JSON Merge Patch: dicts merge recursively, ``null`` deletes
+    a key, everything else replaces."""
+    if not isinstance(patch, dict):
+        return patch
+    if not isinstance(target, dict):
+        target = {}
+    out = dict(target)
+    for key, value in patch.items():
+        if value is None:
+            out.pop(key, None)
+        else:
+            out[key] = _merge_patch(out.get(key), value)
+    return out
+
+
 def _status_body(code: int, reason: str, message: str) -> dict:
     """metav1.Status — the exact error body a kube-apiserver returns, so
     any k8s client library can consume this server's errors."""
@@ -364,6 +380,44 @@ def build_app(client: Client, token: str = None) -> FastAPI:
         except Exception as exc:
             return _error_response(exc)
 
+    async def patch_object(plural: str, name: str, request: Request):
+        """RFC 7386 JSON Merge Patch (``Content-Type:
+        application/merge-patch+json`` — what ``kubectl patch`` sends by
+        default for custom resources): fetch, merge (null deletes), write
+        back through the normal admission/validation chain.  The
+        read-merge-write carries the stored resourceVersion and retries
+        briefly on concurrent-writer conflicts, as the apiserver does."""
+        cls = _cls(plural)
+        ctype = request.headers.get("content-type", "")
+        if "merge-patch" not in ctype and "application/json" not in ctype:
+            return _error_response(ValueError(
+                f"unsupported patch content-type {ctype!r} "
+                "(application/merge-patch+json)"))
+        try:
+            patch = await request.json()
+        except Exception as exc:
+            return _error_response(ValueError(f"unreadable patch body: {exc}"))
+        status_sub = name.endswith("/status")
+        if status_sub:
+            name = name[: -len("/status")]
+        last: Exception = None
+        for _ in range(8):
+            try:
+                current = client.get(cls, name)
+                merged = _merge_patch(current.model_dump(by_alias=True), patch)
+                obj = cls.model_validate(merged)
+                obj.metadata.name = name
+                # carry the rv we read so a concurrent writer conflicts
+                obj.metadata.resourceVersion = current.metadata.resourceVersion
+                if status_sub:
+                    return _dump(client.update_status(obj))
+                return _dump(client.update(obj))
+            except ConflictError as exc:
+                last = exc
+            except Exception as exc:
+                return _error_response(exc)
+        return _error_response(last)
+
     def delete_object(plural: str, name: str):
         cls = _cls(plural)
         try:
@@ -391,6 +445,7 @@ def build_app(client: Client, token: str = None) -> FastAPI:
         app.get(base + "/{plural}/{name:path}")(get_object)
         app.post(base + "/{plural}", status_code=201)(create_object)
         app.put(base + "/{plural}/{name:path}")(update_object)
+        app.patch(base + "/{plural}/{name:path}")(patch_object)
         app.delete(base + "/{plural}/{name:path}")(delete_object)
 
     return app

@@ -183,3 +183,17 @@ def test_get_with_label_selector(cli, tmp_path):
         "get", "composableresources", "-l", "app.kubernetes.io/managed-by=ghost"
     )
     assert rc == 0 and "gpu-" not in out
+
+
+def test_patch_command(cli):
+    run, stack = cli
+    from tests.conftest import make_request
+
+    stack.mgr.client.create(make_request("pz", target_node="node0"))
+    rc, out, _ = run("patch", "composabilityrequests", "pz",
+                     "-p", '{"spec": {"resource": {"size": 4}}}')
+    assert rc == 0 and "patched" in out
+    from cro_amd.api.v1alpha1.types import ComposabilityRequest
+
+    assert stack.mgr.client.get(
+        ComposabilityRequest, "pz").spec.resource.size == 4

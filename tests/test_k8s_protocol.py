@@ -336,3 +336,60 @@ def test_field_selector_metadata_name(stack):
     )
     assert resp.status_code == 422
     assert resp.json()["kind"] == "Status"
+
+
+def test_merge_patch_rfc7386(stack):
+    """PATCH with application/merge-patch+json (kubectl patch default for
+    CRs): recursive merge, null deletes, unknown content-type → 422,
+    conflicts retried server-side; the status subresource patches only
+    status."""
+    http, mgr = stack
+    mgr.client.create(make_request("p1", target_node="n0"))
+
+    # spec field merge
+    resp = http.patch(
+        f"{GROUP}/composabilityrequests/p1",
+        json={"spec": {"resource": {"size": 5}}},
+        headers={"Content-Type": "application/merge-patch+json"},
+    )
+    assert resp.status_code == 200, resp.text
+    body = resp.json()
+    assert body["spec"]["resource"]["size"] == 5
+    assert body["spec"]["resource"]["model"] == "mi355x"  # untouched
+
+    # null deletes a key (resets to schema default on validate)
+    resp = http.patch(
+        f"{GROUP}/composabilityrequests/p1",
+        json={"metadata": {"labels": None}},
+        headers={"Content-Type": "application/merge-patch+json"},
+    )
+    assert resp.status_code == 200
+    assert resp.json()["metadata"]["labels"] == {}
+
+    # status subresource patch touches only status
+    resp = http.patch(
+        f"{GROUP}/composabilityrequests/p1/status",
+        json={"status": {"state": "NodeAllocating"}},
+        headers={"Content-Type": "application/merge-patch+json"},
+    )
+    assert resp.status_code == 200
+    got = mgr.client.get(ComposabilityRequest, "p1")
+    assert got.status.state == "NodeAllocating"
+    assert got.spec.resource.size == 5  # spec untouched by status patch
+
+    # wrong content type → 422 Status
+    resp = http.patch(
+        f"{GROUP}/composabilityrequests/p1",
+        content=b"not json",
+        headers={"Content-Type": "application/strategic-merge-patch+json"},
+    )
+    assert resp.status_code == 422
+    assert resp.json()["kind"] == "Status"
+
+    # schema still enforced through the patch path
+    resp = http.patch(
+        f"{GROUP}/composabilityrequests/p1",
+        json={"spec": {"resource": {"type": "not-a-type"}}},
+        headers={"Content-Type": "application/merge-patch+json"},
+    )
+    assert resp.status_code == 422
