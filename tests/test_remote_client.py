@@ -788,3 +788,52 @@ def test_cached_client_relist_prunes_deleted():
     finally:
         _os.environ.pop("CRO_WATCH_TIMEOUT", None)
         rc.close()
+
+
+def test_cached_client_syncer_repairs_drift(api_server):
+    """The upstream syncer over the CACHED client: out-of-band fabric
+    drift (a composed device with no CR) is detected against cached lists
+    and repaired with a detach CR after the grace period — the cache must
+    not hide the CR set from the diff."""
+    from cro_amd.api.v1alpha1.types import ComposableResource
+
+    url, server_mgr = api_server
+    remote = RemoteClient(url, cache=True)
+    fabric = MockFabric(models={"mi355x": 8})
+    mgr = build_manager(
+        Adapter("DRA", fabric), None, client=remote, enable_webhook=False,
+        syncer_period=0.3, syncer_grace=0.5,
+    )
+    ops = MockNodeOps(client=mgr.client)
+    mgr.resource_reconciler.node_ops = ops
+    mgr.syncer.node_ops = ops
+
+    orig_add = fabric.add_resource
+
+    def add_resource(resource):
+        did, cdi = orig_add(resource)
+        ops.fabric_composed(resource.spec.target_node, did)
+        return did, cdi
+
+    fabric.add_resource = add_resource
+    node = Node()
+    node.metadata.name = "drift-node"
+    node.status.capacity.milli_cpu = 64000
+    node.status.capacity.memory = 1 << 40
+    node.status.capacity.allowed_pod_number = 128
+    server_mgr.client.create(node)
+    mgr.start()
+    try:
+        # out-of-band: the fabric composes a device behind the operator's back
+        did = next(iter(fabric._pool))
+        fabric.force_attach(did, "drift-node")
+        ops.fabric_composed("drift-node", did)
+        # the syncer (reading CRs through the cache) must repair: a
+        # ready-to-detach CR appears, drains, and the fabric releases it
+        assert wait_for(lambda: fabric.attached_to("drift-node") == [], timeout=30), \
+            fabric.attached_to("drift-node")
+        assert wait_for(
+            lambda: server_mgr.client.list(ComposableResource) == [], timeout=30)
+    finally:
+        mgr.stop()
+        remote.close()

@@ -93,3 +93,34 @@ def test_samenode_update_resolves_incoming_implicit_target(client):
     # different node → no collision
     elsewhere = make_request("elsewhere", size=1, target_node="node8")
     assert validate_composability_request(mine, [elsewhere]) is None
+
+
+def test_terminating_requests_still_conflict(guarded_client):
+    """Reference rule scope: the webhook lists ALL requests — a request
+    mid-deletion (deletionTimestamp set, finalizer pending) still counts
+    for duplicate detection (composabilityrequest_webhook.go:85-89 lists
+    without filtering)."""
+    from cro_amd.api.v1alpha1.types import ComposabilityRequest
+
+    guarded_client.create(make_request("dying", target_node="nodeA"))
+    got = guarded_client.get(ComposabilityRequest, "dying")
+    got.metadata.finalizers.append("com.ie.ibm.hpsys/finalizer")
+    guarded_client.update(got)
+    guarded_client.delete(ComposabilityRequest, "dying")  # deletionTimestamp set
+    assert guarded_client.get(
+        ComposabilityRequest, "dying").metadata.deletionTimestamp is not None
+
+    with pytest.raises(AdmissionDenied, match="already exists"):
+        guarded_client.create(make_request("newer", target_node="nodeA"))
+
+
+def test_differentnode_duplicate_across_terminating(guarded_client):
+    from cro_amd.api.v1alpha1.types import ComposabilityRequest
+
+    guarded_client.create(make_request("d1", policy="differentnode"))
+    got = guarded_client.get(ComposabilityRequest, "d1")
+    got.metadata.finalizers.append("com.ie.ibm.hpsys/finalizer")
+    guarded_client.update(got)
+    guarded_client.delete(ComposabilityRequest, "d1")
+    with pytest.raises(AdmissionDenied, match="already exists"):
+        guarded_client.create(make_request("d2", policy="differentnode"))
